@@ -92,7 +92,17 @@ class GraphItem:
         self._models: List[torch.nn.Module] = []
         self._savers: list = []          # checkpoint.Saver registrations
         self._param_to_name: Dict[int, str] = {}
+        self._recorded_modules: list = []    # raw scope captures (patch.py)
+        self._recorded_optimizers: list = []
         self._prepared = False
+
+    # -- deferred scope capture (reference patch.py interception) ----------
+    def record_module(self, module: torch.nn.Module):
+        """Record a module constructed under scope(); resolved in prepare()."""
+        self._recorded_modules.append(module)
+
+    def record_optimizer(self, optimizer: torch.optim.Optimizer):
+        self._recorded_optimizers.append(optimizer)
 
     # -- capture ----------------------------------------------------------
     def extend_model(self, model: torch.nn.Module, name_prefix: str = ""):
@@ -147,8 +157,26 @@ class GraphItem:
         self._savers.append(saver)
 
     def prepare(self):
-        """Harvest any optimizer params not yet named (reference prepare,
+        """Resolve scope captures: register top-level recorded modules (those
+        not contained in another recorded module), then optimizers — so every
+        optimizer param gets its qualified module name (reference prepare,
         graph_item.py:494-497)."""
+        if self._recorded_modules:
+            contained = set()
+            for m in self._recorded_modules:
+                for sub in m.modules():
+                    if sub is not m:
+                        contained.add(id(sub))
+            seen = set()
+            for m in self._recorded_modules:
+                if id(m) in contained or id(m) in seen:
+                    continue
+                seen.add(id(m))
+                self.extend_model(m)
+            self._recorded_modules.clear()
+        for opt in self._recorded_optimizers:
+            self.extend_optimizer_info(opt)
+        self._recorded_optimizers.clear()
         self._prepared = True
         return self
 

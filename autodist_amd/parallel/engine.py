@@ -1,0 +1,607 @@
+"""DistributedEngine — the per-rank execution engine.
+
+This is the MI355X-native replacement for the reference's graph
+transformation + TF distributed runtime. Where the reference rewrites a TF
+GraphDef (GraphTransformer.transform, kernel/graph_transformer.py:55-92) and
+lets TF's C++ executor run inserted CollectiveReduce / accumulator / queue
+ops, this engine runs one process per GPU and installs the synchronization
+directly:
+
+  * AllReduce vars   -> flat gradient buckets + RCCL all-reduce on a comm
+                        HIP stream, issued from post-accumulate-grad hooks so
+                        collectives overlap backward
+                        (reference: all_reduce_synchronizer.py:102-130).
+  * PS vars          -> shard owners hold a master copy + shard-local
+                        optimizer state; reduce->apply->broadcast pipelined on
+                        the comm stream with optional bounded staleness
+                        (reference: ps_synchronizer.py:250-458).
+  * Partitioned vars -> per-shard owners/buckets (partitioner.py semantics).
+  * Sparse vars      -> variable-length allgather + segment-coalesce + rowwise
+                        apply (reference: all_reduce_synchronizer.py:132-173,
+                        ps_synchronizer.py:476-535).
+
+The engine applies optimizer updates itself (parallel/apply.py) — required
+for shard-local state — matching torch.optim numerics exactly.
+"""
+import dataclasses
+import os
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from autodist_amd.const import DEFAULT_BUCKET_BYTES, DEFAULT_MASTER_ADDR, \
+    DEFAULT_MASTER_PORT
+from autodist_amd.parallel import apply as apply_mod
+from autodist_amd.parallel.buckets import Bucket, build_buckets
+from autodist_amd.parallel.comm import allgather_sparse, coalesce_rows
+from autodist_amd.parallel.compressor import Compressor
+from autodist_amd.parallel.partitioner import ShardSlice, make_shard_slices
+from autodist_amd.proto.strategy_ir import CompressorType
+from autodist_amd.utils import logging
+
+_ASYNC_PS_MAX_DEPTH = 4  # queue cap for sync=False (unbounded staleness)
+
+
+@dataclasses.dataclass
+class ShardPlan:
+    name: str
+    kind: str                      # "allreduce" | "ps"
+    slice: Optional[ShardSlice]    # None = whole variable
+    group: int = 0                 # AR bucket group
+    owner_rank: int = 0            # PS owner
+    compressor: CompressorType = CompressorType.NoneCompressor
+    sync: bool = True
+    staleness: int = 0
+    local_replication: bool = False
+    # runtime state
+    state: Optional[dict] = None           # optimizer state (applier ranks)
+    master: Optional[torch.Tensor] = None  # PS master copy (owner)
+    stage: Optional[torch.Tensor] = None   # PS broadcast staging buffer
+    reducer: Optional["ShardReducer"] = None
+
+
+@dataclasses.dataclass
+class VarPlan:
+    name: str
+    param: torch.nn.Parameter
+    sparse: bool
+    cls_name: str                  # optimizer class
+    hyper: dict                    # optimizer hyperparams for this param
+    shards: List[ShardPlan] = dataclasses.field(default_factory=list)
+    bucketed: bool = False
+
+
+class ShardReducer:
+    """Direct all-reduce of one gradient shard (partitioned-AR path)."""
+
+    def __init__(self, plan: ShardPlan, param: torch.nn.Parameter):
+        self.plan = plan
+        self.param = param
+        self.compressor = Compressor.create(plan.compressor, plan.name)
+        self._buf: Optional[torch.Tensor] = None
+        self._handle = None
+        self._issued = False
+        self._is_view = False
+
+    def reset(self):
+        self._handle = None
+        self._issued = False
+
+    def _shard_grad(self):
+        g = self.param.grad
+        sl = self.plan.slice
+        if sl is None:
+            view = g
+        else:
+            view = sl.view(g)
+        if view.is_contiguous():
+            self._is_view = True
+            return view
+        self._is_view = False
+        if self._buf is None or self._buf.shape != view.shape:
+            self._buf = torch.empty_like(view, memory_format=torch.contiguous_format)
+        self._buf.copy_(view)
+        return self._buf
+
+    def issue(self, engine):
+        if self._issued:
+            return
+        self._issued = True
+        if engine.world_size <= 1:
+            return
+        if engine.device.type == "cuda":
+            ev = torch.cuda.Event()
+            ev.record()
+            engine.comm_stream.wait_event(ev)
+            with torch.cuda.stream(engine.comm_stream):
+                self._reduce(engine)
+        else:
+            self._reduce(engine)
+
+    def _reduce(self, engine):
+        t = self._shard_grad()
+        t.mul_(1.0 / engine.world_size)
+        self._handle = self.compressor.reduce(t, group=engine.process_group,
+                                              async_op=True)
+        self._reduced_tensor = t
+
+    def finalize(self, engine):
+        if engine.world_size <= 1 or not self._issued:
+            return
+        if engine.device.type == "cuda":
+            with torch.cuda.stream(engine.comm_stream):
+                self.compressor.finalize(self._reduced_tensor, self._handle)
+                if not self._is_view:
+                    self.plan.slice.view(self.param.grad).copy_(self._reduced_tensor)
+            ev = torch.cuda.Event()
+            ev.record(engine.comm_stream)
+            torch.cuda.current_stream().wait_event(ev)
+        else:
+            self.compressor.finalize(self._reduced_tensor, self._handle)
+            if not self._is_view:
+                self.plan.slice.view(self.param.grad).copy_(self._reduced_tensor)
+
+
+class _PSRound:
+    """One in-flight PS reduce->apply->broadcast round (staleness pipeline)."""
+
+    def __init__(self, shard: ShardPlan, event=None, handle=None):
+        self.shard = shard
+        self.event = event     # GPU: comm-stream event after broadcast
+        self.handle = handle   # CPU: async work handle of broadcast
+
+
+class DistributedEngine:
+    """One rank's engine. Builds plans from a compiled Strategy and runs the
+    per-step synchronization + update protocol."""
+
+    def __init__(self, graph_item, strategy, rank: Optional[int] = None,
+                 world_size: Optional[int] = None,
+                 device: Optional[torch.device] = None,
+                 bucket_bytes: int = DEFAULT_BUCKET_BYTES,
+                 process_group=None):
+        self.graph_item = graph_item
+        self.strategy = strategy
+        self.rank = rank if rank is not None else int(os.environ.get("RANK", 0))
+        self.world_size = world_size if world_size is not None else int(
+            os.environ.get("WORLD_SIZE", 1))
+        self.process_group = process_group
+        if device is not None:
+            self.device = device
+        elif torch.cuda.is_available():
+            local = int(os.environ.get("LOCAL_RANK", self.rank % max(
+                torch.cuda.device_count(), 1)))
+            self.device = torch.device("cuda", local)
+        else:
+            self.device = torch.device("cpu")
+        self.bucket_bytes = bucket_bytes
+        self.comm_stream = None
+        self.buckets: List[Bucket] = []
+        self.var_plans: List[VarPlan] = []
+        self._hook_handles = []
+        self._ps_outstanding: Dict[str, List[_PSRound]] = {}
+        self._replica_ranks: Dict[str, int] = {}
+        self._step_count = 0
+        self._fallback_user_opt = False
+        self._setup_done = False
+        # RCCL supports ReduceOp.AVG; we pre-divide instead so one code path
+        # covers gloo + compressors (the divide is fused into the HIP pack /
+        # cast kernels on GPU).
+        self.avg_supported = False
+        self.avg_supported_needs_post_div = False
+
+    # ------------------------------------------------------------------ set-up
+    def maybe_init_process_group(self):
+        if self.world_size <= 1 or dist.is_initialized():
+            return
+        backend = "nccl" if self.device.type == "cuda" else "gloo"
+        if self.device.type == "cuda":
+            torch.cuda.set_device(self.device)
+        addr = os.environ.get("MASTER_ADDR", DEFAULT_MASTER_ADDR)
+        port = os.environ.get("MASTER_PORT", str(DEFAULT_MASTER_PORT))
+        dist.init_process_group(
+            backend=backend, init_method=f"tcp://{addr}:{port}",
+            rank=self.rank, world_size=self.world_size)
+        logging.info("process group ready: backend=%s rank=%d world=%d",
+                     backend, self.rank, self.world_size)
+
+    def setup(self):
+        if self._setup_done:
+            return self
+        self.maybe_init_process_group()
+        if self.device.type == "cuda":
+            torch.cuda.set_device(self.device)
+            self.comm_stream = torch.cuda.Stream(device=self.device)
+        for m in self.graph_item.models:
+            m.to(self.device)
+        self._map_replica_ranks()
+        self._build_plans()
+        self._sync_initial_params()
+        self._build_buckets_and_hooks()
+        self._patch_optimizer()
+        self._setup_done = True
+        return self
+
+    def _map_replica_ranks(self):
+        for i, dev in enumerate(self.strategy.graph_config.replicas):
+            self._replica_ranks[dev] = i
+
+    def _owner_rank_of(self, destination: str) -> int:
+        if destination in self._replica_ranks:
+            return self._replica_ranks[destination]
+        # CPU destination or unknown device: host rank 0 owns it
+        return 0
+
+    def _hyper_for(self, param) -> tuple:
+        opt_item = self.graph_item.optimizers[0] if self.graph_item.optimizers else None
+        if opt_item is None:
+            raise RuntimeError("no optimizer captured — build one inside scope()")
+        opt = opt_item.optimizer
+        if opt is not None:
+            for group in opt.param_groups:
+                for p in group["params"]:
+                    if p is param:
+                        return opt_item.cls_name, {
+                            k: v for k, v in group.items() if k != "params"}
+        return opt_item.cls_name, dict(opt_item.defaults)
+
+    def _build_plans(self):
+        vars_by_name = self.graph_item.trainable_var_op_to_var
+        for node in self.strategy.node_config:
+            item = vars_by_name.get(node.var_name)
+            if item is None or item.param is None:
+                continue
+            cls_name, hyper = self._hyper_for(item.param)
+            plan = VarPlan(name=node.var_name, param=item.param,
+                           sparse=item.is_sparse, cls_name=cls_name, hyper=hyper)
+            if node.part_config:
+                slices = make_shard_slices(item.shape, node.partitioner)
+                for sl, part in zip(slices, node.part_config):
+                    plan.shards.append(self._make_shard(part, sl))
+            else:
+                plan.shards.append(self._make_shard(node, None))
+            self.var_plans.append(plan)
+        known = {p.name for p in self.var_plans}
+        missing = [n for n in vars_by_name if n not in known
+                   and vars_by_name[n].param is not None]
+        if missing:
+            logging.warning("vars without strategy config (treated as "
+                            "local-only): %s", missing[:5])
+        unsupported = {p.cls_name for p in self.var_plans
+                       if not apply_mod.is_supported(p.cls_name)}
+        if unsupported:
+            all_simple_ar = all(
+                len(p.shards) == 1 and p.shards[0].kind == "allreduce"
+                and not p.sparse for p in self.var_plans)
+            if not all_simple_ar:
+                raise NotImplementedError(
+                    f"optimizer(s) {unsupported} not supported by the engine "
+                    "applier; PS/partitioned/sparse strategies need one of "
+                    f"{('SGD', 'Adam', 'AdamW', 'Adagrad', 'RMSprop')}")
+            self._fallback_user_opt = True
+            logging.info("unsupported optimizer %s: falling back to user "
+                         "optimizer.step() after gradient sync", unsupported)
+
+    def _make_shard(self, node, sl: Optional[ShardSlice]) -> ShardPlan:
+        if node.all_reduce_synchronizer is not None:
+            s = node.all_reduce_synchronizer
+            return ShardPlan(name=node.var_name, kind="allreduce", slice=sl,
+                             group=s.group, compressor=s.compressor)
+        if node.ps_synchronizer is not None:
+            s = node.ps_synchronizer
+            return ShardPlan(name=node.var_name, kind="ps", slice=sl,
+                             owner_rank=self._owner_rank_of(s.reduction_destination),
+                             sync=s.sync, staleness=s.staleness,
+                             local_replication=s.local_replication)
+        # no synchronizer: local-only (treated as AR group 0 w/o collective)
+        return ShardPlan(name=node.var_name, kind="allreduce", slice=sl)
+
+    def _sync_initial_params(self):
+        if self.world_size <= 1:
+            return
+        for plan in self.var_plans:
+            dist.broadcast(plan.param.data, src=0, group=self.process_group)
+
+    def _build_buckets_and_hooks(self):
+        bucket_items = []
+        hooked = set()
+        for plan in self.var_plans:
+            if plan.sparse:
+                plan.param.grad = None
+                continue
+            whole = len(plan.shards) == 1 and plan.shards[0].slice is None
+            sh0 = plan.shards[0]
+            if whole and sh0.kind == "allreduce":
+                plan.bucketed = True
+                bucket_items.append((plan.param, sh0.group, sh0.compressor))
+            elif sh0.kind == "allreduce":
+                # partitioned AR: per-shard direct reducers
+                for sh in plan.shards:
+                    sh.reducer = ShardReducer(sh, plan.param)
+                if id(plan.param) not in hooked:
+                    hooked.add(id(plan.param))
+                    self._hook_handles.append(
+                        plan.param.register_post_accumulate_grad_hook(
+                            self._make_shard_hook(plan)))
+            else:
+                # PS shards: allocate master/stage on owner/all ranks
+                for sh in plan.shards:
+                    view = sh.slice.view(plan.param.data) if sh.slice \
+                        else plan.param.data
+                    sh.stage = view.detach().clone().contiguous()
+                    if self.rank == sh.owner_rank:
+                        sh.master = view.detach().clone().contiguous()
+                        sh.state = apply_mod.make_state(
+                            plan.cls_name, sh.master, plan.hyper)
+                    self._ps_outstanding[sh.name + f"/{sh.slice.start if sh.slice else 0}"] = []
+        # PS/partitioned params keep ordinary grads; bucketed params get views
+        self.buckets = build_buckets(bucket_items, self.device, self.bucket_bytes)
+        param_to_bucket = {}
+        for b in self.buckets:
+            for p in b.params:
+                param_to_bucket[id(p)] = b
+        for plan in self.var_plans:
+            if plan.bucketed:
+                b = param_to_bucket[id(plan.param)]
+                if id(plan.param) not in hooked:
+                    hooked.add(id(plan.param))
+                    self._hook_handles.append(
+                        plan.param.register_post_accumulate_grad_hook(
+                            self._make_bucket_hook(b)))
+                # AR whole-var optimizer state on every rank
+                sh = plan.shards[0]
+                if sh.state is None and not self._fallback_user_opt:
+                    sh.state = apply_mod.make_state(
+                        plan.cls_name, plan.param, plan.hyper)
+            elif plan.shards[0].kind == "allreduce" and plan.shards[0].reducer:
+                for sh in plan.shards:
+                    if sh.state is None:
+                        ref = sh.slice.view(plan.param.data) if sh.slice \
+                            else plan.param.data
+                        sh.state = apply_mod.make_state(plan.cls_name, ref,
+                                                        plan.hyper)
+
+    def _make_bucket_hook(self, bucket: Bucket):
+        def hook(_param):
+            bucket.mark_ready_and_maybe_issue(self)
+        return hook
+
+    def _make_shard_hook(self, plan: VarPlan):
+        def hook(_param):
+            for sh in plan.shards:
+                sh.reducer.issue(self)
+        return hook
+
+    def _patch_optimizer(self):
+        """Route the captured optimizer's step/zero_grad through the engine
+        (the torch-idiom analog of the reference's apply_gradients patch,
+        autodist/patch.py:79-88)."""
+        opt = self.graph_item.optimizer
+        if opt is None:
+            return
+        engine = self
+
+        def step(closure=None):  # noqa: ARG001
+            engine.step()
+
+        def zero_grad(set_to_none=True):  # noqa: ARG001
+            engine.zero_grad()
+
+        opt._autodist_orig_step = opt.step
+        opt.step = step
+        opt._autodist_orig_zero_grad = opt.zero_grad
+        opt.zero_grad = zero_grad
+
+    # ------------------------------------------------------------------ steps
+    def zero_grad(self):
+        for b in self.buckets:
+            b.zero_()
+            b.reset()
+        for plan in self.var_plans:
+            if not plan.bucketed:
+                plan.param.grad = None
+                for sh in plan.shards:
+                    if sh.reducer is not None:
+                        sh.reducer.reset()
+
+    def step(self):
+        """Synchronize gradients + apply updates. Call after backward."""
+        # 1) flush collectives not yet issued (e.g. params w/o grads)
+        for b in self.buckets:
+            if not b._issued:
+                b.issue(self)
+        for plan in self.var_plans:
+            for sh in plan.shards:
+                if sh.reducer is not None:
+                    sh.reducer.issue(self)
+        # 2) finalize AR (compute stream now depends on reduced grads)
+        for b in self.buckets:
+            b.finalize(self)
+        for plan in self.var_plans:
+            for sh in plan.shards:
+                if sh.reducer is not None:
+                    sh.reducer.finalize(self)
+        # 3) sparse path
+        for plan in self.var_plans:
+            if plan.sparse:
+                self._sync_and_apply_sparse(plan)
+        # 4) PS rounds
+        for plan in self.var_plans:
+            if plan.sparse:
+                continue
+            for sh in plan.shards:
+                if sh.kind == "ps":
+                    self._issue_ps_round(plan, sh)
+        # 5) dense applies (AR vars) — grouped multi-tensor
+        if self._fallback_user_opt:
+            opt = self.graph_item.optimizer
+            orig = getattr(opt, "_autodist_orig_step", opt.step)
+            orig()
+        else:
+            self._apply_dense_updates()
+        # 6) consume due PS rounds (staleness bound)
+        for plan in self.var_plans:
+            if plan.sparse:
+                continue
+            for sh in plan.shards:
+                if sh.kind == "ps":
+                    self._consume_ps_rounds(plan, sh)
+        self._step_count += 1
+
+    # -- dense AR apply ----------------------------------------------------
+    def _apply_dense_updates(self):
+        groups: Dict[tuple, list] = {}
+        for plan in self.var_plans:
+            if plan.sparse:
+                continue
+            for sh in plan.shards:
+                if sh.kind != "allreduce":
+                    continue
+                if sh.slice is None:
+                    p, g = plan.param.data, plan.param.grad
+                else:
+                    p = sh.slice.view(plan.param.data)
+                    g = sh.slice.view(plan.param.grad)
+                key = (plan.cls_name, _hyper_key(plan.hyper))
+                groups.setdefault(key, []).append((p, g, sh.state, plan.hyper))
+        for (cls_name, _), items in groups.items():
+            params = [x[0] for x in items]
+            grads = [x[1] for x in items]
+            states = [x[2] for x in items]
+            apply_mod.apply_dense(cls_name, params, grads, states, items[0][3])
+
+    # -- PS path -----------------------------------------------------------
+    def _ps_key(self, sh: ShardPlan) -> str:
+        return sh.name + f"/{sh.slice.start if sh.slice else 0}"
+
+    def _issue_ps_round(self, plan: VarPlan, sh: ShardPlan):
+        grad = plan.param.grad
+        if grad is None:
+            return
+        gview = sh.slice.view(grad) if sh.slice else grad
+        gbuf = gview if gview.is_contiguous() else gview.contiguous()
+
+        def round_body():
+            gbuf.mul_(1.0 / self.world_size)
+            dist.reduce(gbuf, dst=sh.owner_rank, group=self.process_group)
+            if self.rank == sh.owner_rank:
+                apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
+                                      [sh.state], plan.hyper)
+                src_buf = sh.master
+            else:
+                src_buf = sh.stage
+            dist.broadcast(src_buf, src=sh.owner_rank, group=self.process_group)
+            if self.rank == sh.owner_rank:
+                sh.stage.copy_(sh.master)
+
+        if self.world_size <= 1:
+            # degenerate single-rank PS: apply to master, stage it
+            gbuf2 = gbuf
+            apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf2],
+                                  [sh.state], plan.hyper)
+            sh.stage.copy_(sh.master)
+            self._ps_outstanding[self._ps_key(sh)].append(_PSRound(sh))
+            return
+        if self.device.type == "cuda":
+            ev = torch.cuda.Event()
+            ev.record()
+            self.comm_stream.wait_event(ev)
+            with torch.cuda.stream(self.comm_stream):
+                round_body()
+            done = torch.cuda.Event()
+            done.record(self.comm_stream)
+            self._ps_outstanding[self._ps_key(sh)].append(_PSRound(sh, event=done))
+        else:
+            round_body()
+            self._ps_outstanding[self._ps_key(sh)].append(_PSRound(sh))
+
+    def _consume_ps_rounds(self, plan: VarPlan, sh: ShardPlan):
+        """Pop rounds older than the staleness bound and install their
+        results into the live parameter (reference staleness queues,
+        ps_synchronizer.py:388-458; sync barrier 335-385)."""
+        key = self._ps_key(sh)
+        rounds = self._ps_outstanding[key]
+        depth = sh.staleness if sh.sync else _ASYNC_PS_MAX_DEPTH
+        while len(rounds) > depth:
+            r = rounds.pop(0)
+            if r.event is not None:
+                torch.cuda.current_stream().wait_event(r.event)
+            if r.handle is not None:
+                r.handle.wait()
+            view = sh.slice.view(plan.param.data) if sh.slice else plan.param.data
+            view.copy_(sh.stage)
+
+    def drain(self):
+        """Consume ALL outstanding PS rounds (end of training / checkpoint)."""
+        for plan in self.var_plans:
+            if plan.sparse:
+                continue
+            for sh in plan.shards:
+                if sh.kind == "ps":
+                    key = self._ps_key(sh)
+                    rounds = self._ps_outstanding.get(key, [])
+                    while rounds:
+                        r = rounds.pop(0)
+                        if r.event is not None:
+                            torch.cuda.current_stream().wait_event(r.event)
+                    view = sh.slice.view(plan.param.data) if sh.slice \
+                        else plan.param.data
+                    view.copy_(sh.stage)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)
+
+    # -- sparse path ---------------------------------------------------------
+    def _sync_and_apply_sparse(self, plan: VarPlan):
+        grad = plan.param.grad
+        if grad is None:
+            return
+        if grad.is_sparse:
+            grad = grad.coalesce()
+            indices, values = grad.indices()[0], grad.values()
+        else:
+            # dense grad on a sparse-flagged var: treat all rows as touched
+            indices = torch.arange(grad.shape[0], device=grad.device)
+            values = grad
+        if self.world_size > 1:
+            indices, values = allgather_sparse(
+                indices, values, self.world_size, self.process_group)
+        indices, values = coalesce_rows(indices, values)
+        values = values / float(self.world_size)
+        # replicated rowwise apply: identical on every rank == PS result
+        sh = plan.shards[0]
+        if sh.state is None:
+            sh.state = apply_mod.make_state(plan.cls_name, plan.param.data,
+                                            plan.hyper)
+        cls = plan.cls_name
+        if cls == "Adam" and plan.hyper.get("_sparse_adam"):
+            cls = "SparseAdam"
+        apply_mod.apply_sparse_rows(cls, plan.param.data, indices, values,
+                                    sh.state, plan.hyper)
+
+    # ------------------------------------------------------------------ misc
+    @property
+    def step_count(self) -> int:
+        return self._step_count
+
+    def teardown(self):
+        for h in self._hook_handles:
+            h.remove()
+        self._hook_handles.clear()
+        opt = self.graph_item.optimizer
+        if opt is not None and hasattr(opt, "_autodist_orig_step"):
+            opt.step = opt._autodist_orig_step
+        if opt is not None and hasattr(opt, "_autodist_orig_zero_grad"):
+            opt.zero_grad = opt._autodist_orig_zero_grad
+
+
+def _hyper_key(hyper: dict) -> tuple:
+    out = []
+    for k in sorted(hyper):
+        v = hyper[k]
+        if isinstance(v, (list, tuple)):
+            v = tuple(v)
+        elif not isinstance(v, (int, float, bool, str, type(None))):
+            v = str(v)
+        out.append((k, v))
+    return tuple(out)
