@@ -63,6 +63,72 @@ def ext():
 
 # -- kernel wrappers (HIP on GPU, torch fallback on CPU) --------------------
 
+def fused_apply(cls_name: str, param: torch.Tensor, grad: torch.Tensor,
+                state: dict, hyper: dict) -> bool:
+    """One-kernel optimizer update over flat bucket buffers. Returns False
+    when this optimizer class has no fused kernel (caller falls back to the
+    torch implementation)."""
+    if not (param.is_cuda and has_gpu_ops()):
+        return False
+    e = ext()
+    if cls_name == "SGD":
+        momentum = float(hyper.get("momentum", 0.0))
+        buf = None
+        first = False
+        if momentum != 0.0:
+            if "momentum_buffer" not in state:
+                state["momentum_buffer"] = torch.empty_like(param)
+                first = True
+            buf = state["momentum_buffer"]
+        e.fused_sgd(param, grad, buf, float(hyper["lr"]), momentum,
+                    float(hyper.get("dampening", 0.0)),
+                    float(hyper.get("weight_decay", 0.0)),
+                    bool(hyper.get("nesterov", False)), first,
+                    bool(hyper.get("maximize", False)))
+        return True
+    if cls_name in ("Adam", "AdamW"):
+        state["step"] += 1
+        step = float(state["step"])
+        beta1, beta2 = hyper.get("betas", (0.9, 0.999))
+        bc1 = 1.0 - beta1 ** step
+        sqrt_bc2 = (1.0 - beta2 ** step) ** 0.5
+        wd_default = 0.01 if cls_name == "AdamW" else 0.0
+        e.fused_adam(param, grad, state["exp_avg"], state["exp_avg_sq"],
+                     float(hyper["lr"]), float(beta1), float(beta2),
+                     float(hyper.get("eps", 1e-8)),
+                     float(hyper.get("weight_decay", wd_default)),
+                     cls_name == "AdamW", bc1, sqrt_bc2,
+                     bool(hyper.get("maximize", False)))
+        return True
+    return False
+
+
+def scale_cast_bf16(inp: torch.Tensor, out: torch.Tensor, scale: float):
+    if inp.is_cuda and has_gpu_ops():
+        ext().scale_cast_bf16(inp, out, float(scale))
+    else:
+        out.copy_(inp if scale == 1.0 else inp * scale)
+
+
+def cast_back_f32(wire: torch.Tensor, out: torch.Tensor):
+    if wire.is_cuda and has_gpu_ops():
+        ext().cast_back_f32(wire, out)
+    else:
+        out.copy_(wire)
+
+
+def ef_compress(flat: torch.Tensor, err: torch.Tensor, wire: torch.Tensor,
+                scale: float):
+    """flat += err; wire = bf16(flat*scale); err = flat - fp32(wire)/scale."""
+    if flat.is_cuda and has_gpu_ops():
+        ext().ef_compress(flat, err, wire, float(scale))
+    else:
+        flat.add_(err)
+        torch.mul(flat, scale, out=err)       # reuse err as tmp
+        wire.copy_(err)
+        err.copy_(flat).sub_(wire.to(torch.float32) / scale)
+
+
 def segment_coalesce(indices: torch.Tensor, values: torch.Tensor):
     """Sort-by-index + segment-sum of duplicate rows (sparse accumulator)."""
     if indices.is_cuda and has_gpu_ops():

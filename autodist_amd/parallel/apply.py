@@ -190,6 +190,17 @@ def apply_dense(cls_name: str, params, grads, states, hyper):
     _APPLY[cls_name](params, grads, states, hyper)
 
 
+def apply_flat(cls_name: str, param: torch.Tensor, grad: torch.Tensor,
+               state: dict, hyper: dict):
+    """Bucket-level update over flat buffers: ONE hand-written gfx950 HIP
+    kernel launch on GPU (ops/csrc/multi_tensor.hip), torch fallback on CPU."""
+    from autodist_amd.ops import api as ops_api
+    if param.is_cuda and ops_api.has_gpu_ops():
+        if ops_api.fused_apply(cls_name, param, grad, state, hyper):
+            return
+    apply_dense(cls_name, [param], [grad], [state], hyper)
+
+
 # -- sparse (row-wise) applies: the SparseApply* table ----------------------
 
 def apply_sparse_rows(cls_name: str, param: torch.Tensor, rows: torch.Tensor,

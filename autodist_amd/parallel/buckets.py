@@ -20,19 +20,28 @@ from autodist_amd.utils import logging
 
 
 class Bucket:
-    """One fusion group: a flat buffer + member params whose .grad are views."""
+    """One fusion group: flat grad + flat param (+ flat optimizer state)
+    buffers; member params and their .grad are VIEWS into the flats.
+
+    Flattening params too means the whole bucket's optimizer update is ONE
+    elementwise HIP kernel over contiguous HBM3E buffers — no multi-tensor
+    metadata, perfectly coalesced 16 B/lane access."""
 
     def __init__(self, bucket_id: int, dtype: torch.dtype, device: torch.device,
-                 compressor: Compressor):
+                 compressor: Compressor, cls_name: str = "", hyper: dict = None):
         self.id = bucket_id
         self.dtype = dtype
         self.device = device
         self.compressor = compressor
+        self.cls_name = cls_name          # optimizer class for this bucket
+        self.hyper = hyper or {}
         self.params: List[torch.nn.Parameter] = []
         self.shapes: List[torch.Size] = []
         self.offsets: List[int] = []
         self.numel = 0
-        self.flat: Optional[torch.Tensor] = None
+        self.flat: Optional[torch.Tensor] = None        # gradients
+        self.flat_param: Optional[torch.Tensor] = None  # parameter values
+        self.state: dict = {}                           # flat optimizer state
         self._ready = 0
         self._handle = None
         self.done_event: Optional[torch.cuda.Event] = None
@@ -45,10 +54,17 @@ class Bucket:
         self.numel += param.numel()
 
     def allocate(self):
-        """Allocate the flat buffer and point every member's .grad at a view."""
+        """Allocate flat buffers; repoint every member's .data and .grad at
+        views of them (values preserved)."""
         self.flat = torch.zeros(self.numel, dtype=self.dtype, device=self.device)
-        for p, off, shape in zip(self.params, self.offsets, self.shapes):
-            p.grad = self.flat[off:off + p.numel()].view(shape)
+        self.flat_param = torch.empty(self.numel, dtype=self.dtype,
+                                      device=self.device)
+        with torch.no_grad():
+            for p, off, shape in zip(self.params, self.offsets, self.shapes):
+                n = p.numel()
+                self.flat_param[off:off + n].view(shape).copy_(p.data)
+                p.data = self.flat_param[off:off + n].view(shape)
+                p.grad = self.flat[off:off + n].view(shape)
 
     @property
     def nbytes(self) -> int:
@@ -89,10 +105,9 @@ class Bucket:
             self._reduce(engine)
 
     def _reduce(self, engine):
-        if not engine.avg_supported:
-            self.flat.mul_(1.0 / engine.world_size)
         self._handle = self.compressor.reduce(
-            self.flat, group=engine.process_group, async_op=True)
+            self.flat, group=engine.process_group, async_op=True,
+            scale=1.0 / engine.world_size)
 
     def finalize(self, engine):
         """Make the compute stream depend on this bucket's reduced result."""
@@ -114,29 +129,34 @@ class Bucket:
 
 def build_buckets(items, device: torch.device,
                   bucket_bytes: int = DEFAULT_BUCKET_BYTES) -> List[Bucket]:
-    """Group (param, group_id, compressor_type) triples into Buckets.
+    """Group (param, group_id, compressor_type, cls_name, hyper, hyper_key)
+    tuples into Buckets.
 
     Keeps the strategy's group ids (one ScopedAllocator-group == >=1 buckets),
     splitting any group larger than bucket_bytes so collectives overlap with
-    backward instead of waiting for one giant buffer.
+    backward instead of waiting for one giant buffer. Buckets are homogeneous
+    in (dtype, optimizer class, hyperparams) so the whole bucket updates with
+    one fused kernel.
 
     Bucket order follows REVERSED registration order within each group, since
     autograd produces gradients roughly in reverse forward order — the first
     bucket to fill is the one holding the last layers.
     """
     by_group: Dict[tuple, list] = {}
-    for param, group_id, comp_type in items:
-        by_group.setdefault((group_id, param.dtype), []).append((param, comp_type))
+    for param, group_id, comp_type, cls_name, hyper, hyper_key in items:
+        key = (group_id, param.dtype, cls_name, hyper_key)
+        by_group.setdefault(key, []).append((param, comp_type, hyper))
     buckets: List[Bucket] = []
-    for (group_id, dtype), members in sorted(by_group.items(),
-                                             key=lambda kv: kv[0][0]):
+    for (group_id, dtype, cls_name, _), members in sorted(
+            by_group.items(), key=lambda kv: (kv[0][0], kv[0][2], kv[0][3])):
         members = list(reversed(members))
         current = None
         elt = torch.empty((), dtype=dtype).element_size()
-        for param, comp_type in members:
+        for param, comp_type, hyper in members:
             if current is None or current.numel * elt >= bucket_bytes:
                 comp = Compressor.create(comp_type, f"bucket{len(buckets)}")
-                current = Bucket(len(buckets), dtype, device, comp)
+                current = Bucket(len(buckets), dtype, device, comp,
+                                 cls_name=cls_name, hyper=hyper)
                 buckets.append(current)
             current.add(param)
         logging.debug("group %s -> %d bucket(s)", group_id, len(buckets))

@@ -40,10 +40,11 @@ class Compressor:
             return PowerSGDCompressor(var_name, **kwargs)
         raise ValueError(f"unknown compressor {kind}")
 
-    def reduce(self, flat: torch.Tensor, group=None,
-               async_op: bool = False):
-        """All-reduce (mean) the flat gradient tensor in place; returns a
-        handle when async (reference _all_reduce, compressor.py:84-96)."""
+    def reduce(self, flat: torch.Tensor, group=None, async_op: bool = False,
+               scale: float = 1.0):
+        """All-reduce the flat gradient tensor; `scale` (typically 1/world)
+        is FUSED into the compress kernel so the mean costs no extra pass
+        (reference _all_reduce, compressor.py:84-96). Returns an async handle."""
         raise NotImplementedError
 
     def finalize(self, flat: torch.Tensor, handle) -> None:
@@ -55,7 +56,9 @@ class Compressor:
 class NoneCompressor(Compressor):
     """Pass-through all-reduce (reference compressor.py:146-166)."""
 
-    def reduce(self, flat, group=None, async_op=False):
+    def reduce(self, flat, group=None, async_op=False, scale=1.0):
+        if scale != 1.0:
+            flat.mul_(scale)
         handle = dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=group,
                                  async_op=async_op)
         return handle
@@ -64,7 +67,8 @@ class NoneCompressor(Compressor):
 class HorovodCompressor(Compressor):
     """Cast-compress to bf16 on the wire (reference compressor.py:169-201;
     the reference casts fp32->fp16, we use bf16: same 2x wire saving, no
-    range loss, native CDNA4 dtype)."""
+    range loss, native CDNA4 dtype). The scale+cast is one fused gfx950
+    kernel (ops/csrc/multi_tensor.hip scale_cast_bf16_kernel)."""
 
     WIRE_DTYPE = torch.bfloat16
 
@@ -72,48 +76,43 @@ class HorovodCompressor(Compressor):
         super().__init__(var_name)
         self._wire: Optional[torch.Tensor] = None
 
-    def _compress(self, flat: torch.Tensor) -> torch.Tensor:
-        if flat.dtype == self.WIRE_DTYPE:
-            return flat
+    def _wire_buf(self, flat: torch.Tensor) -> torch.Tensor:
         if self._wire is None or self._wire.numel() != flat.numel() \
                 or self._wire.device != flat.device:
             self._wire = torch.empty_like(flat, dtype=self.WIRE_DTYPE)
-        self._wire.copy_(flat)
         return self._wire
 
-    def reduce(self, flat, group=None, async_op=False):
-        wire = self._compress(flat)
+    def reduce(self, flat, group=None, async_op=False, scale=1.0):
+        from autodist_amd.ops import api as ops_api
+        wire = self._wire_buf(flat)
+        ops_api.scale_cast_bf16(flat, wire, scale)
         handle = dist.all_reduce(wire, op=dist.ReduceOp.SUM, group=group,
                                  async_op=async_op)
         return (handle, wire)
 
     def finalize(self, flat, handle) -> None:
+        from autodist_amd.ops import api as ops_api
         h, wire = handle
         if h is not None:
             h.wait()
-        if wire is not flat:
-            flat.copy_(wire)
+        ops_api.cast_back_f32(wire, flat)
 
 
 class HorovodCompressorEF(HorovodCompressor):
     """Cast compression with error feedback (reference compressor.py:204-205):
-    error = flat - decompress(compress(flat)) is added back next step."""
+    error = flat - decompress(compress(flat)) is added back next step.
+    accumulate+cast+error-update run as ONE fused kernel on gfx950."""
 
     def __init__(self, var_name: str = ""):
         super().__init__(var_name)
         self._error: Optional[torch.Tensor] = None
 
-    def reduce(self, flat, group=None, async_op=False):
-        if self._error is not None:
-            flat.add_(self._error)
-        else:
+    def reduce(self, flat, group=None, async_op=False, scale=1.0):
+        from autodist_amd.ops import api as ops_api
+        if self._error is None:
             self._error = torch.zeros_like(flat)
-        wire = self._compress(flat)
-        # error = flat - wire (in flat dtype), BEFORE the collective mutates wire
-        if wire is not flat:
-            self._error.copy_(flat).sub_(wire.to(flat.dtype))
-        else:
-            self._error.zero_()
+        wire = self._wire_buf(flat)
+        ops_api.ef_compress(flat, self._error, wire, scale)
         handle = dist.all_reduce(wire, op=dist.ReduceOp.SUM, group=group,
                                  async_op=async_op)
         return (handle, wire)

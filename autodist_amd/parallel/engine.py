@@ -121,9 +121,9 @@ class ShardReducer:
 
     def _reduce(self, engine):
         t = self._shard_grad()
-        t.mul_(1.0 / engine.world_size)
         self._handle = self.compressor.reduce(t, group=engine.process_group,
-                                              async_op=True)
+                                              async_op=True,
+                                              scale=1.0 / engine.world_size)
         self._reduced_tensor = t
 
     def finalize(self, engine):
@@ -314,7 +314,9 @@ class DistributedEngine:
             sh0 = plan.shards[0]
             if whole and sh0.kind == "allreduce":
                 plan.bucketed = True
-                bucket_items.append((plan.param, sh0.group, sh0.compressor))
+                bucket_items.append((plan.param, sh0.group, sh0.compressor,
+                                     plan.cls_name, plan.hyper,
+                                     _hyper_key(plan.hyper)))
             elif sh0.kind == "allreduce":
                 # partitioned AR: per-shard direct reducers
                 for sh in plan.shards:
@@ -349,11 +351,6 @@ class DistributedEngine:
                     self._hook_handles.append(
                         plan.param.register_post_accumulate_grad_hook(
                             self._make_bucket_hook(b)))
-                # AR whole-var optimizer state on every rank
-                sh = plan.shards[0]
-                if sh.state is None and not self._fallback_user_opt:
-                    sh.state = apply_mod.make_state(
-                        plan.cls_name, plan.param, plan.hyper)
             elif plan.shards[0].kind == "allreduce" and plan.shards[0].reducer:
                 for sh in plan.shards:
                     if sh.state is None:
@@ -451,9 +448,18 @@ class DistributedEngine:
 
     # -- dense AR apply ----------------------------------------------------
     def _apply_dense_updates(self):
+        # bucketed params: one fused flat update per bucket
+        for b in self.buckets:
+            if not b.params:
+                continue
+            if not b.state:
+                b.state = apply_mod.make_state(b.cls_name, b.flat_param, b.hyper)
+            apply_mod.apply_flat(b.cls_name, b.flat_param, b.flat, b.state,
+                                 b.hyper)
+        # non-bucketed AR shards (partitioned): grouped multi-tensor
         groups: Dict[tuple, list] = {}
         for plan in self.var_plans:
-            if plan.sparse:
+            if plan.sparse or plan.bucketed:
                 continue
             for sh in plan.shards:
                 if sh.kind != "allreduce":
