@@ -6,11 +6,200 @@
 #include <ATen/hip/HIPContext.h>
 
 #include "multi_tensor.hip"
+#include "bn_ops.hip"
 
 namespace {
 
 inline hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+// ---------------------------------------------------------------- fused BN
+struct BNGeom {
+  long M;
+  int C;
+  int tx_count, rows_per_blk, grid_c;
+};
+
+BNGeom bn_geom(const torch::Tensor& x) {
+  TORCH_CHECK(x.dim() == 4, "BN expects NCHW-logical channels_last tensor");
+  int C = (int)x.size(1);
+  TORCH_CHECK(C % BN_VEC == 0, "C must be a multiple of 8, got ", C);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x must be channels_last");
+  BNGeom g;
+  g.M = x.size(0) * x.size(2) * x.size(3);
+  g.C = C;
+  g.tx_count = std::min(C / BN_VEC, BLOCK_THREADS);
+  g.rows_per_blk = BLOCK_THREADS / g.tx_count;
+  g.grid_c = (C / BN_VEC + g.tx_count - 1) / g.tx_count;
+  return g;
+}
+
+inline int bn_grid_m(const BNGeom& g, long target_blocks) {
+  long gm = (g.M + g.rows_per_blk - 1) / g.rows_per_blk;
+  long cap = std::max<long>(target_blocks / std::max(g.grid_c, 1), 1);
+  return (int)std::min(gm, cap);
+}
+
+std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
+                                        torch::Tensor bias,
+                                        torch::Tensor running_mean,
+                                        torch::Tensor running_var,
+                                        c10::optional<torch::Tensor> res,
+                                        double eps, double momentum,
+                                        bool relu) {
+  auto g = bn_geom(x);
+  auto fopt = weight.options().dtype(torch::kFloat32);
+  auto sum = torch::zeros({g.C}, fopt);
+  auto sumsq = torch::zeros({g.C}, fopt);
+  auto save_mean = torch::empty({g.C}, fopt);
+  auto save_rstd = torch::empty({g.C}, fopt);
+  auto scale = torch::empty({g.C}, fopt);
+  auto shift = torch::empty({g.C}, fopt);
+  auto y = torch::empty_like(x);
+  dim3 block(BLOCK_THREADS);
+  dim3 grid_r(bn_grid_m(g, 1024), g.grid_c);
+  dim3 grid_a(bn_grid_m(g, 4096), g.grid_c);
+  auto st = cur_stream();
+
+#define BN_FWD_T(T, GET)                                                      \
+  {                                                                           \
+    hipLaunchKernelGGL((bn_fwd_reduce_kernel<T>), grid_r, block, 0, st,       \
+                       GET(x), sum.data_ptr<float>(), sumsq.data_ptr<float>(),\
+                       g.M, g.C);                                             \
+    hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((g.C + 255) / 256),       \
+                       dim3(256), 0, st, sum.data_ptr<float>(),               \
+                       sumsq.data_ptr<float>(), weight.data_ptr<float>(),     \
+                       bias.data_ptr<float>(), running_mean.data_ptr<float>(),\
+                       running_var.data_ptr<float>(),                         \
+                       save_mean.data_ptr<float>(),                           \
+                       save_rstd.data_ptr<float>(), scale.data_ptr<float>(),  \
+                       shift.data_ptr<float>(), g.M, g.C, (float)eps,         \
+                       (float)momentum);                                      \
+    if (res.has_value()) {                                                    \
+      if (relu)                                                               \
+        hipLaunchKernelGGL((bn_fwd_apply_kernel<T, true, true>), grid_a,      \
+                           block, 0, st, GET(x), GET(*res), GET(y),           \
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),  \
+                           g.M, g.C);                                         \
+      else                                                                    \
+        hipLaunchKernelGGL((bn_fwd_apply_kernel<T, false, true>), grid_a,     \
+                           block, 0, st, GET(x), GET(*res), GET(y),           \
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),  \
+                           g.M, g.C);                                         \
+    } else {                                                                  \
+      if (relu)                                                               \
+        hipLaunchKernelGGL((bn_fwd_apply_kernel<T, true, false>), grid_a,     \
+                           block, 0, st, GET(x), (const T*)nullptr, GET(y),   \
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),  \
+                           g.M, g.C);                                         \
+      else                                                                    \
+        hipLaunchKernelGGL((bn_fwd_apply_kernel<T, false, false>), grid_a,    \
+                           block, 0, st, GET(x), (const T*)nullptr, GET(y),   \
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),  \
+                           g.M, g.C);                                         \
+    }                                                                         \
+  }
+
+#define GET_BF16(t) reinterpret_cast<__hip_bfloat16*>((t).data_ptr())
+#define GET_F32(t) (t).data_ptr<float>()
+  if (x.scalar_type() == torch::kBFloat16) {
+    BN_FWD_T(__hip_bfloat16, GET_BF16)
+  } else {
+    TORCH_CHECK(x.scalar_type() == torch::kFloat32);
+    BN_FWD_T(float, GET_F32)
+  }
+  return {y, save_mean, save_rstd};
+}
+
+std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
+                                  torch::Tensor y, torch::Tensor save_mean,
+                                  torch::Tensor save_rstd,
+                                  torch::Tensor weight, bool relu,
+                                  bool need_dres) {
+  auto g = bn_geom(x);
+  auto fopt = weight.options().dtype(torch::kFloat32);
+  auto sum_dz = torch::zeros({g.C}, fopt);
+  auto sum_dzxh = torch::zeros({g.C}, fopt);
+  auto k1 = torch::empty({g.C}, fopt);
+  auto k2 = torch::empty({g.C}, fopt);
+  auto k3 = torch::empty({g.C}, fopt);
+  auto dweight = torch::empty({g.C}, fopt);
+  auto dbias = torch::empty({g.C}, fopt);
+  auto dx = torch::empty_like(x);
+  torch::Tensor dres;
+  if (need_dres) dres = torch::empty_like(x);
+  dim3 block(BLOCK_THREADS);
+  dim3 grid_r(bn_grid_m(g, 1024), g.grid_c);
+  dim3 grid_a(bn_grid_m(g, 4096), g.grid_c);
+  auto st = cur_stream();
+
+#define BN_BWD_T(T, GET)                                                      \
+  {                                                                           \
+    if (relu)                                                                 \
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), grid_r, block, 0,   \
+                         st, GET(x), GET(dy), GET(y),                         \
+                         save_mean.data_ptr<float>(),                         \
+                         save_rstd.data_ptr<float>(),                         \
+                         sum_dz.data_ptr<float>(),                            \
+                         sum_dzxh.data_ptr<float>(), g.M, g.C);               \
+    else                                                                      \
+      hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid_r, block, 0,  \
+                         st, GET(x), GET(dy), GET(y),                         \
+                         save_mean.data_ptr<float>(),                         \
+                         save_rstd.data_ptr<float>(),                         \
+                         sum_dz.data_ptr<float>(),                            \
+                         sum_dzxh.data_ptr<float>(), g.M, g.C);               \
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((g.C + 255) / 256),       \
+                       dim3(256), 0, st, sum_dz.data_ptr<float>(),            \
+                       sum_dzxh.data_ptr<float>(), weight.data_ptr<float>(),  \
+                       save_rstd.data_ptr<float>(), k1.data_ptr<float>(),     \
+                       k2.data_ptr<float>(), k3.data_ptr<float>(),            \
+                       dweight.data_ptr<float>(), dbias.data_ptr<float>(),    \
+                       g.M, g.C);                                             \
+    T* dres_p = need_dres ? GET(dres) : (T*)nullptr;                          \
+    if (relu) {                                                               \
+      if (need_dres)                                                          \
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, true, true>), grid_a,      \
+                           block, 0, st, GET(x), GET(dy), GET(y),             \
+                           save_mean.data_ptr<float>(),                       \
+                           save_rstd.data_ptr<float>(), k1.data_ptr<float>(), \
+                           k2.data_ptr<float>(), k3.data_ptr<float>(),        \
+                           GET(dx), dres_p, g.M, g.C);                        \
+      else                                                                    \
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, true, false>), grid_a,     \
+                           block, 0, st, GET(x), GET(dy), GET(y),             \
+                           save_mean.data_ptr<float>(),                       \
+                           save_rstd.data_ptr<float>(), k1.data_ptr<float>(), \
+                           k2.data_ptr<float>(), k3.data_ptr<float>(),        \
+                           GET(dx), dres_p, g.M, g.C);                        \
+    } else {                                                                  \
+      if (need_dres)                                                          \
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, false, true>), grid_a,     \
+                           block, 0, st, GET(x), GET(dy), GET(y),             \
+                           save_mean.data_ptr<float>(),                       \
+                           save_rstd.data_ptr<float>(), k1.data_ptr<float>(), \
+                           k2.data_ptr<float>(), k3.data_ptr<float>(),        \
+                           GET(dx), dres_p, g.M, g.C);                        \
+      else                                                                    \
+        hipLaunchKernelGGL((bn_bwd_apply_kernel<T, false, false>), grid_a,    \
+                           block, 0, st, GET(x), GET(dy), GET(y),             \
+                           save_mean.data_ptr<float>(),                       \
+                           save_rstd.data_ptr<float>(), k1.data_ptr<float>(), \
+                           k2.data_ptr<float>(), k3.data_ptr<float>(),        \
+                           GET(dx), dres_p, g.M, g.C);                        \
+    }                                                                         \
+  }
+  if (x.scalar_type() == torch::kBFloat16) {
+    BN_BWD_T(__hip_bfloat16, GET_BF16)
+  } else {
+    TORCH_CHECK(x.scalar_type() == torch::kFloat32);
+    BN_BWD_T(float, GET_F32)
+  }
+  std::vector<torch::Tensor> out = {dx, dweight, dbias};
+  if (need_dres) out.push_back(dres);
+  return out;
 }
 
 void check_f32_flat(const torch::Tensor& t, const char* name) {
@@ -156,4 +345,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dedup-sum row-sparse gradient");
   m.def("scatter_add_rows", &scatter_add_rows, "rowwise scatter-add");
   m.def("gather_rows", &gather_rows, "rowwise gather");
+  m.def("bn_fwd_train", &bn_fwd_train,
+        "fused NHWC batchnorm fwd (+add+relu), returns y/save_mean/save_rstd");
+  m.def("bn_bwd", &bn_bwd,
+        "fused NHWC batchnorm bwd (+relu-mask+dres), returns "
+        "dx/dweight/dbias[/dres]");
 }

@@ -31,6 +31,8 @@ def parse_args():
     p.add_argument("--bucket-mb", type=int, default=25)
     p.add_argument("--image-size", type=int, default=224)
     p.add_argument("--no-channels-last", action="store_true")
+    p.add_argument("--no-fused-bn", action="store_true",
+                   help="disable the hand-written gfx950 fused BN kernels")
     return p.parse_args()
 
 
@@ -70,7 +72,8 @@ def main():
     from autodist_amd.resource_spec import ResourceSpec
 
     torch.manual_seed(1234)
-    model = getattr(resnet, args.model)(num_classes=1000)
+    fused = use_cuda and not args.no_fused_bn
+    model = getattr(resnet, args.model)(num_classes=1000, fused=fused)
     model = model.to(device)
     channels_last = use_cuda and not args.no_channels_last
     if channels_last:

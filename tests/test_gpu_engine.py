@@ -19,7 +19,7 @@ def test_resnet50_train_step_bf16():
         memory_format=torch.channels_last)
     g = GraphItem()
     g.extend_model(model)
-    opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+    opt = torch.optim.SGD(model.parameters(), lr=0.005, momentum=0.9)
     g.extend_optimizer_info(opt)
     engine = DistributedEngine(g, AllReduce().build(g, ResourceSpec()),
                                rank=0, world_size=1, device=device).setup()
@@ -33,7 +33,7 @@ def test_resnet50_train_step_bf16():
             loss = torch.nn.functional.cross_entropy(model(x), y)
         loss.backward()
         opt.step()
-        losses.append(float(loss))
+        losses.append(loss.item())
     torch.cuda.synchronize()
     assert losses[-1] < losses[0]
     assert all(v == v for v in losses)

@@ -121,6 +121,79 @@ def test_gather_scatter_rows(ext):
     assert torch.allclose(acc, ref)
 
 
+def test_fused_bn_gpu_vs_cpu_reference(ext):
+    """HIP fused BN (bf16 NHWC, +add+relu) vs the CPU fp32 torch reference."""
+    torch.manual_seed(0)
+    N, C, H, W = 8, 64, 14, 14
+    x32 = torch.randn(N, C, H, W)
+    res32 = torch.randn(N, C, H, W)
+    w = torch.rand(C) + 0.5
+    b = torch.randn(C)
+    # CPU fp32 reference
+    rm_c, rv_c = torch.zeros(C), torch.ones(C)
+    xc = x32.clone().requires_grad_(True)
+    rc = res32.clone().requires_grad_(True)
+    wc = torch.nn.Parameter(w.clone())
+    bc = torch.nn.Parameter(b.clone())
+    y_ref = torch.relu(torch.nn.functional.batch_norm(
+        xc, rm_c, rv_c, wc, bc, training=True, momentum=0.1, eps=1e-5) + rc)
+    gy = torch.randn(N, C, H, W)
+    y_ref.backward(gy)
+    # GPU bf16 fused
+    from autodist_amd.ops.fused_bn import fused_bn_train
+    dev = torch.device("cuda")
+    xg = x32.to(dev, torch.bfloat16).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    rg = res32.to(dev, torch.bfloat16).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    wg = torch.nn.Parameter(w.to(dev))
+    bg = torch.nn.Parameter(b.to(dev))
+    rm_g, rv_g = torch.zeros(C, device=dev), torch.ones(C, device=dev)
+    y = fused_bn_train(xg, wg, bg, rm_g, rv_g, momentum=0.1, eps=1e-5,
+                       relu=True, residual=rg)
+    y.backward(gy.to(dev, torch.bfloat16).contiguous(
+        memory_format=torch.channels_last))
+    tol = 5e-2  # bf16 inputs
+    assert torch.allclose(y.detach().float().cpu(), y_ref.detach(), atol=tol)
+    assert torch.allclose(rm_g.cpu(), rm_c, atol=1e-2)
+    assert torch.allclose(rv_g.cpu(), rv_c, atol=1e-2)
+    assert torch.allclose(xg.grad.float().cpu(), xc.grad, atol=tol)
+    assert torch.allclose(rg.grad.float().cpu(), rc.grad, atol=tol)
+    assert torch.allclose(wg.grad.cpu(), wc.grad, atol=0.5), \
+        (wg.grad.cpu() - wc.grad).abs().max()
+    assert torch.allclose(bg.grad.cpu(), bc.grad, atol=0.5)
+
+
+def test_fused_bn_gpu_fp32_exact(ext):
+    """fp32 path of the HIP BN kernels: tight tolerance vs torch."""
+    torch.manual_seed(2)
+    N, C, H, W = 4, 32, 7, 7
+    dev = torch.device("cuda")
+    x32 = torch.randn(N, C, H, W, device=dev)
+    w = torch.nn.Parameter(torch.rand(C, device=dev) + 0.5)
+    b = torch.nn.Parameter(torch.randn(C, device=dev))
+    rm, rv = torch.zeros(C, device=dev), torch.ones(C, device=dev)
+    from autodist_amd.ops.fused_bn import fused_bn_train
+    xg = x32.clone().contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    y = fused_bn_train(xg, w, b, rm, rv, relu=True)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xc = x32.clone().cpu().requires_grad_(True)
+    wc = torch.nn.Parameter(w.detach().cpu())
+    bc = torch.nn.Parameter(b.detach().cpu())
+    rmc, rvc = torch.zeros(C), torch.ones(C)
+    yr = torch.relu(torch.nn.functional.batch_norm(
+        xc, rmc, rvc, wc, bc, training=True, momentum=0.1, eps=1e-5))
+    yr.backward(gy.cpu())
+    assert torch.allclose(y.detach().cpu(), yr.detach(), atol=1e-4)
+    assert torch.allclose(xg.grad.cpu(), xc.grad, atol=1e-4)
+    assert torch.allclose(w.grad.cpu(), wc.grad, atol=1e-2)
+    assert torch.allclose(b.grad.cpu(), bc.grad, atol=1e-2)
+    assert torch.allclose(rm.cpu(), rmc, atol=1e-5)
+    assert torch.allclose(rv.cpu(), rvc, atol=1e-4)
+
+
 def test_apply_flat_dispatch_uses_hip():
     """apply_flat on GPU must route through the HIP kernel and match the
     CPU torch reference."""
