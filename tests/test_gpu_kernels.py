@@ -135,8 +135,9 @@ def test_fused_bn_gpu_vs_cpu_reference(ext):
     rc = res32.clone().requires_grad_(True)
     wc = torch.nn.Parameter(w.clone())
     bc = torch.nn.Parameter(b.clone())
-    y_ref = torch.relu(torch.nn.functional.batch_norm(
-        xc, rm_c, rv_c, wc, bc, training=True, momentum=0.1, eps=1e-5) + rc)
+    z_ref = torch.nn.functional.batch_norm(
+        xc, rm_c, rv_c, wc, bc, training=True, momentum=0.1, eps=1e-5) + rc
+    y_ref = torch.relu(z_ref)
     gy = torch.randn(N, C, H, W)
     y_ref.backward(gy)
     # GPU bf16 fused
@@ -157,8 +158,13 @@ def test_fused_bn_gpu_vs_cpu_reference(ext):
     assert torch.allclose(y.detach().float().cpu(), y_ref.detach(), atol=tol)
     assert torch.allclose(rm_g.cpu(), rm_c, atol=1e-2)
     assert torch.allclose(rv_g.cpu(), rv_c, atol=1e-2)
-    assert torch.allclose(xg.grad.float().cpu(), xc.grad, atol=tol)
-    assert torch.allclose(rg.grad.float().cpu(), rc.grad, atol=tol)
+    # dx/dres: near the ReLU boundary (|y|~bf16 eps) the fp32 and bf16 masks
+    # legitimately disagree; compare away from the boundary.
+    interior = z_ref.detach().abs() > 1e-2
+    dxe = (xg.grad.float().cpu() - xc.grad).abs()
+    assert dxe[interior].max() < tol, dxe[interior].max()
+    dre = (rg.grad.float().cpu() - rc.grad).abs()
+    assert dre[interior].max() < tol, dre[interior].max()
     assert torch.allclose(wg.grad.cpu(), wc.grad, atol=0.5), \
         (wg.grad.cpu() - wc.grad).abs().max()
     assert torch.allclose(bg.grad.cpu(), bc.grad, atol=0.5)
