@@ -90,8 +90,10 @@ __global__ void bn_fwd_reduce_kernel(const T* __restrict__ x,
 
 // ------------------------------------------------------------ fwd finalize
 // scale = w*rstd; shift = b - mean*scale; running stats updated in place.
-__global__ void bn_fwd_finalize_kernel(const float* __restrict__ sum,
-                                       const float* __restrict__ sumsq,
+// NOTE: consumes and RE-ZEROES sum/sumsq so the workspace buffers can be
+// reused without a fill kernel (they are allocated zeroed once per module).
+__global__ void bn_fwd_finalize_kernel(float* __restrict__ sum,
+                                       float* __restrict__ sumsq,
                                        const float* __restrict__ weight,
                                        const float* __restrict__ bias,
                                        float* __restrict__ running_mean,
@@ -105,6 +107,8 @@ __global__ void bn_fwd_finalize_kernel(const float* __restrict__ sum,
   if (c >= C) return;
   float mean = sum[c] / (float)M;
   float var = fmaxf(sumsq[c] / (float)M - mean * mean, 0.f);
+  sum[c] = 0.f;
+  sumsq[c] = 0.f;
   float rstd = rsqrtf(var + eps);
   float sc = weight[c] * rstd;
   save_mean[c] = mean;
@@ -218,8 +222,9 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
 // ------------------------------------------------------------ bwd finalize
 // k1 = w*rstd ; k2 = sum_dz/M ; k3 = sum_dzxh/M ; dweight = sum_dzxh ;
 // dbias = sum_dz
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_dz,
-                                       const float* __restrict__ sum_dzxh,
+// consumes and RE-ZEROES sum_dz/sum_dzxh (see fwd finalize note)
+__global__ void bn_bwd_finalize_kernel(float* __restrict__ sum_dz,
+                                       float* __restrict__ sum_dzxh,
                                        const float* __restrict__ weight,
                                        const float* __restrict__ save_rstd,
                                        float* __restrict__ k1,
@@ -230,11 +235,14 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_dz,
                                        int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  float sdz = sum_dz[c], sdzxh = sum_dzxh[c];
+  sum_dz[c] = 0.f;
+  sum_dzxh[c] = 0.f;
   k1[c] = weight[c] * save_rstd[c];
-  k2[c] = sum_dz[c] / (float)M;
-  k3[c] = sum_dzxh[c] / (float)M;
-  dweight[c] = sum_dzxh[c];
-  dbias[c] = sum_dz[c];
+  k2[c] = sdz / (float)M;
+  k3[c] = sdzxh / (float)M;
+  dweight[c] = sdzxh;
+  dbias[c] = sdz;
 }
 
 // -------------------------------------------------------------- bwd apply

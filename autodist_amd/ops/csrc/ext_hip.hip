@@ -48,15 +48,22 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         torch::Tensor running_var,
                                         c10::optional<torch::Tensor> res,
                                         double eps, double momentum,
-                                        bool relu) {
+                                        bool relu,
+                                        c10::optional<torch::Tensor> ws) {
+  // ws: persistent per-module workspace [6, C] fp32, rows:
+  //   0 sum (zeroed), 1 sumsq (zeroed), 2 save_mean, 3 save_rstd,
+  //   4 scale, 5 shift. Finalize re-zeroes rows 0-1 after consuming them.
   auto g = bn_geom(x);
   auto fopt = weight.options().dtype(torch::kFloat32);
-  auto sum = torch::zeros({g.C}, fopt);
-  auto sumsq = torch::zeros({g.C}, fopt);
-  auto save_mean = torch::empty({g.C}, fopt);
-  auto save_rstd = torch::empty({g.C}, fopt);
-  auto scale = torch::empty({g.C}, fopt);
-  auto shift = torch::empty({g.C}, fopt);
+  torch::Tensor w6 = ws.has_value() ? *ws
+      : torch::zeros({6, (long)g.C}, fopt);
+  TORCH_CHECK(w6.size(0) >= 6 && w6.size(1) == g.C && w6.is_contiguous());
+  auto sum = w6[0];
+  auto sumsq = w6[1];
+  auto save_mean = w6[2];
+  auto save_rstd = w6[3];
+  auto scale = w6[4];
+  auto shift = w6[5];
   auto y = torch::empty_like(x);
   dim3 block(BLOCK_THREADS);
   dim3 grid_r(bn_grid_m(g, 1024), g.grid_c);
@@ -117,14 +124,20 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                                   torch::Tensor y, torch::Tensor save_mean,
                                   torch::Tensor save_rstd,
                                   torch::Tensor weight, bool relu,
-                                  bool need_dres) {
+                                  bool need_dres,
+                                  c10::optional<torch::Tensor> ws) {
+  // ws rows: 0 sum_dz (zeroed), 1 sum_dzxh (zeroed), 2 k1, 3 k2, 4 k3.
   auto g = bn_geom(x);
   auto fopt = weight.options().dtype(torch::kFloat32);
-  auto sum_dz = torch::zeros({g.C}, fopt);
-  auto sum_dzxh = torch::zeros({g.C}, fopt);
-  auto k1 = torch::empty({g.C}, fopt);
-  auto k2 = torch::empty({g.C}, fopt);
-  auto k3 = torch::empty({g.C}, fopt);
+  torch::Tensor w5 = ws.has_value() ? *ws
+      : torch::zeros({5, (long)g.C}, fopt);
+  TORCH_CHECK(w5.size(0) >= 5 && w5.size(1) == g.C && w5.is_contiguous());
+  auto sum_dz = w5[0];
+  auto sum_dzxh = w5[1];
+  auto k1 = w5[2];
+  auto k2 = w5[3];
+  auto k3 = w5[4];
+  // fresh allocations: returned to autograd as parameter gradients
   auto dweight = torch::empty({g.C}, fopt);
   auto dbias = torch::empty({g.C}, fopt);
   auto dx = torch::empty_like(x);

@@ -21,12 +21,13 @@ from autodist_amd.ops import api as ops_api
 class _FusedBNFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var,
-                momentum, eps, relu, residual):
+                momentum, eps, relu, residual, ws=None):
         use_hip = x.is_cuda and ops_api.has_gpu_ops()
+        ctx.ws_bwd = ws[1] if ws is not None else None
         if use_hip:
             y, save_mean, save_rstd = ops_api.ext().bn_fwd_train(
                 x, weight, bias, running_mean, running_var, residual,
-                eps, momentum, relu)
+                eps, momentum, relu, ws[0] if ws is not None else None)
         else:
             xf = x.float()
             dims = (0, 2, 3)
@@ -59,7 +60,7 @@ class _FusedBNFunction(torch.autograd.Function):
         if use_hip:
             out = ops_api.ext().bn_bwd(x, dy.contiguous(
                 memory_format=torch.channels_last), y, save_mean, save_rstd,
-                weight, relu, has_res)
+                weight, relu, has_res, ctx.ws_bwd)
             dx, dweight, dbias = out[0], out[1], out[2]
             dres = out[3] if has_res else None
         else:
@@ -79,14 +80,14 @@ class _FusedBNFunction(torch.autograd.Function):
                 - xhat * sum_dzxh[None, :, None, None] / m)
             dx = dxf.to(x.dtype)
             dweight, dbias = sum_dzxh, sum_dz
-        return (dx, dweight, dbias, None, None, None, None, None, dres)
+        return (dx, dweight, dbias, None, None, None, None, None, dres, None)
 
 
 def fused_bn_train(x, weight, bias, running_mean, running_var,
                    momentum=0.1, eps=1e-5, relu=False,
-                   residual: Optional[torch.Tensor] = None):
+                   residual: Optional[torch.Tensor] = None, ws=None):
     return _FusedBNFunction.apply(x, weight, bias, running_mean, running_var,
-                                  momentum, eps, relu, residual)
+                                  momentum, eps, relu, residual, ws)
 
 
 class FusedBatchNorm2d(torch.nn.Module):
@@ -108,6 +109,7 @@ class FusedBatchNorm2d(torch.nn.Module):
         self.register_buffer("running_var", torch.ones(num_features))
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
+        self._ws = None  # persistent kernel workspaces (not in state_dict)
 
     def _check(self, x):
         if x.is_cuda and not x.is_contiguous(memory_format=torch.channels_last):
@@ -120,9 +122,15 @@ class FusedBatchNorm2d(torch.nn.Module):
             residual = self._check(residual)
         if self.training:
             self.num_batches_tracked += 1
+            if x.is_cuda and (self._ws is None
+                              or self._ws[0].device != x.device):
+                self._ws = (
+                    torch.zeros(6, self.num_features, device=x.device),
+                    torch.zeros(5, self.num_features, device=x.device))
             return fused_bn_train(x, self.weight, self.bias,
                                   self.running_mean, self.running_var,
-                                  self.momentum, self.eps, self.relu, residual)
+                                  self.momentum, self.eps, self.relu, residual,
+                                  ws=self._ws if x.is_cuda else None)
         # eval: running-stat normalize (+add+relu)
         scale = self.weight * (self.running_var + self.eps).rsqrt()
         shift = self.bias - self.running_mean * scale

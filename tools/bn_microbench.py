@@ -46,14 +46,17 @@ def main():
         rm, rv = torch.zeros(C, device=dev), torch.ones(C, device=dev)
         gy = torch.randn_like(x)
 
+        ws = (torch.zeros(6, C, device=dev), torch.zeros(5, C, device=dev))
+
         def fused_fwd():
             with torch.no_grad():
-                return fused_bn_train(x, w, b, rm, rv, relu=True, residual=res)
+                return fused_bn_train(x, w, b, rm, rv, relu=True,
+                                      residual=res, ws=ws)
 
         def fused_fwd_bwd():
             xg = x.detach().requires_grad_(True)
             rg = res.detach().requires_grad_(True)
-            y = fused_bn_train(xg, w, b, rm, rv, relu=True, residual=rg)
+            y = fused_bn_train(xg, w, b, rm, rv, relu=True, residual=rg, ws=ws)
             y.backward(gy)
 
         bn = torch.nn.BatchNorm2d(C).to(dev)
