@@ -48,10 +48,15 @@ __device__ __forceinline__ void store8(float* p, const float* v) {
 }
 
 // ---------------------------------------------------------------- fwd reduce
+// Two-stage, atomic-free (deterministic): each block writes one partial row
+// partial[blockIdx.x][c]; the finalize kernel sums the gridM rows. (A global
+// fp32 atomicAdd design serialized ~gridM-way per channel address — ~200 us
+// on small tensors.)
 template <typename T>
 __global__ void bn_fwd_reduce_kernel(const T* __restrict__ x,
-                                     float* __restrict__ sum,
-                                     float* __restrict__ sumsq, long M, int C) {
+                                     float* __restrict__ partial_sum,
+                                     float* __restrict__ partial_sq, long M,
+                                     int C) {
   int tx_count = min(C / BN_VEC, (int)blockDim.x);
   int tx = threadIdx.x % tx_count;
   int ty = threadIdx.x / tx_count;
@@ -80,20 +85,16 @@ __global__ void bn_fwd_reduce_kernel(const T* __restrict__ x,
 #pragma unroll
       for (int k = 0; k < BN_VEC; ++k) { s[k] += ls[o + k]; q[k] += lq[o + k]; }
     }
-#pragma unroll
-    for (int k = 0; k < BN_VEC; ++k) {
-      atomicAdd(sum + c0 + k, s[k]);
-      atomicAdd(sumsq + c0 + k, q[k]);
-    }
+    store8(partial_sum + (long)blockIdx.x * C + c0, s);
+    store8(partial_sq + (long)blockIdx.x * C + c0, q);
   }
 }
 
 // ------------------------------------------------------------ fwd finalize
 // scale = w*rstd; shift = b - mean*scale; running stats updated in place.
-// NOTE: consumes and RE-ZEROES sum/sumsq so the workspace buffers can be
-// reused without a fill kernel (they are allocated zeroed once per module).
-__global__ void bn_fwd_finalize_kernel(float* __restrict__ sum,
-                                       float* __restrict__ sumsq,
+__global__ void bn_fwd_finalize_kernel(const float* __restrict__ partial_sum,
+                                       const float* __restrict__ partial_sq,
+                                       int grid_m,
                                        const float* __restrict__ weight,
                                        const float* __restrict__ bias,
                                        float* __restrict__ running_mean,
@@ -105,10 +106,13 @@ __global__ void bn_fwd_finalize_kernel(float* __restrict__ sum,
                                        int C, float eps, float momentum) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = sum[c] / (float)M;
-  float var = fmaxf(sumsq[c] / (float)M - mean * mean, 0.f);
-  sum[c] = 0.f;
-  sumsq[c] = 0.f;
+  float s = 0.f, q = 0.f;
+  for (int r = 0; r < grid_m; ++r) {
+    s += partial_sum[(long)r * C + c];
+    q += partial_sq[(long)r * C + c];
+  }
+  float mean = s / (float)M;
+  float var = fmaxf(q / (float)M - mean * mean, 0.f);
   float rstd = rsqrtf(var + eps);
   float sc = weight[c] * rstd;
   save_mean[c] = mean;
@@ -167,8 +171,8 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ y,
                                      const float* __restrict__ save_mean,
                                      const float* __restrict__ save_rstd,
-                                     float* __restrict__ sum_dz,
-                                     float* __restrict__ sum_dzxh, long M,
+                                     float* __restrict__ partial_dz,
+                                     float* __restrict__ partial_dzxh, long M,
                                      int C) {
   int tx_count = min(C / BN_VEC, (int)blockDim.x);
   int tx = threadIdx.x % tx_count;
@@ -211,20 +215,17 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
 #pragma unroll
       for (int k = 0; k < BN_VEC; ++k) { s[k] += ls[o + k]; t[k] += lt[o + k]; }
     }
-#pragma unroll
-    for (int k = 0; k < BN_VEC; ++k) {
-      atomicAdd(sum_dz + c0 + k, s[k]);
-      atomicAdd(sum_dzxh + c0 + k, t[k]);
-    }
+    store8(partial_dz + (long)blockIdx.x * C + c0, s);
+    store8(partial_dzxh + (long)blockIdx.x * C + c0, t);
   }
 }
 
 // ------------------------------------------------------------ bwd finalize
 // k1 = w*rstd ; k2 = sum_dz/M ; k3 = sum_dzxh/M ; dweight = sum_dzxh ;
 // dbias = sum_dz
-// consumes and RE-ZEROES sum_dz/sum_dzxh (see fwd finalize note)
-__global__ void bn_bwd_finalize_kernel(float* __restrict__ sum_dz,
-                                       float* __restrict__ sum_dzxh,
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_dz,
+                                       const float* __restrict__ partial_dzxh,
+                                       int grid_m,
                                        const float* __restrict__ weight,
                                        const float* __restrict__ save_rstd,
                                        float* __restrict__ k1,
@@ -235,9 +236,11 @@ __global__ void bn_bwd_finalize_kernel(float* __restrict__ sum_dz,
                                        int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float sdz = sum_dz[c], sdzxh = sum_dzxh[c];
-  sum_dz[c] = 0.f;
-  sum_dzxh[c] = 0.f;
+  float sdz = 0.f, sdzxh = 0.f;
+  for (int r = 0; r < grid_m; ++r) {
+    sdz += partial_dz[(long)r * C + c];
+    sdzxh += partial_dzxh[(long)r * C + c];
+  }
   k1[c] = weight[c] * save_rstd[c];
   k2[c] = sdz / (float)M;
   k3[c] = sdzxh / (float)M;

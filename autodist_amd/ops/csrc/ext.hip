@@ -41,6 +41,9 @@ inline int bn_grid_m(const BNGeom& g, long target_blocks) {
   return (int)std::min(gm, cap);
 }
 
+// max partial rows for the two-stage BN reductions (2 blocks per CU)
+#define BN_GM_MAX 512
+
 std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         torch::Tensor bias,
                                         torch::Tensor running_mean,
@@ -49,34 +52,37 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         double eps, double momentum,
                                         bool relu,
                                         c10::optional<torch::Tensor> ws) {
-  // ws: persistent per-module workspace [6, C] fp32, rows:
-  //   0 sum (zeroed), 1 sumsq (zeroed), 2 save_mean, 3 save_rstd,
-  //   4 scale, 5 shift. Finalize re-zeroes rows 0-1 after consuming them.
+  // ws: persistent per-module workspace [2*BN_GM_MAX + 4, C] fp32:
+  //   rows [0, GM)               partial sums (per reduce block)
+  //   rows [BN_GM_MAX, BN_GM_MAX+GM) partial sumsq
+  //   rows 2*BN_GM_MAX + {0,1,2,3}: save_mean, save_rstd, scale, shift
   auto g = bn_geom(x);
   auto fopt = weight.options().dtype(torch::kFloat32);
   torch::Tensor w6 = ws.has_value() ? *ws
-      : torch::zeros({6, (long)g.C}, fopt);
-  TORCH_CHECK(w6.size(0) >= 6 && w6.size(1) == g.C && w6.is_contiguous());
-  auto sum = w6[0];
-  auto sumsq = w6[1];
-  auto save_mean = w6[2];
-  auto save_rstd = w6[3];
-  auto scale = w6[4];
-  auto shift = w6[5];
+      : torch::empty({2 * BN_GM_MAX + 4, (long)g.C}, fopt);
+  TORCH_CHECK(w6.size(0) >= 2 * BN_GM_MAX + 4 && w6.size(1) == g.C &&
+              w6.is_contiguous());
+  float* wp = w6.data_ptr<float>();
+  float* partial_sum = wp;
+  float* partial_sq = wp + (long)BN_GM_MAX * g.C;
+  auto save_mean = w6[2 * BN_GM_MAX + 0];
+  auto save_rstd = w6[2 * BN_GM_MAX + 1];
+  auto scale = w6[2 * BN_GM_MAX + 2];
+  auto shift = w6[2 * BN_GM_MAX + 3];
   auto y = torch::empty_like(x);
   dim3 block(BLOCK_THREADS);
-  dim3 grid_r(bn_grid_m(g, 1024), g.grid_c);
+  int gm = bn_grid_m(g, BN_GM_MAX);
+  dim3 grid_r(gm, g.grid_c);
   dim3 grid_a(bn_grid_m(g, 4096), g.grid_c);
   auto st = cur_stream();
 
 #define BN_FWD_T(T, GET)                                                      \
   {                                                                           \
     hipLaunchKernelGGL((bn_fwd_reduce_kernel<T>), grid_r, block, 0, st,       \
-                       GET(x), sum.data_ptr<float>(), sumsq.data_ptr<float>(),\
-                       g.M, g.C);                                             \
+                       GET(x), partial_sum, partial_sq, g.M, g.C);            \
     hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((g.C + 255) / 256),       \
-                       dim3(256), 0, st, sum.data_ptr<float>(),               \
-                       sumsq.data_ptr<float>(), weight.data_ptr<float>(),     \
+                       dim3(256), 0, st, partial_sum, partial_sq, gm,         \
+                       weight.data_ptr<float>(),                              \
                        bias.data_ptr<float>(), running_mean.data_ptr<float>(),\
                        running_var.data_ptr<float>(),                         \
                        save_mean.data_ptr<float>(),                           \
@@ -125,17 +131,20 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                                   torch::Tensor weight, bool relu,
                                   bool need_dres,
                                   c10::optional<torch::Tensor> ws) {
-  // ws rows: 0 sum_dz (zeroed), 1 sum_dzxh (zeroed), 2 k1, 3 k2, 4 k3.
+  // ws rows: [0,GM) partial_dz, [BN_GM_MAX, BN_GM_MAX+GM) partial_dzxh,
+  //          2*BN_GM_MAX + {0,1,2}: k1, k2, k3.
   auto g = bn_geom(x);
   auto fopt = weight.options().dtype(torch::kFloat32);
   torch::Tensor w5 = ws.has_value() ? *ws
-      : torch::zeros({5, (long)g.C}, fopt);
-  TORCH_CHECK(w5.size(0) >= 5 && w5.size(1) == g.C && w5.is_contiguous());
-  auto sum_dz = w5[0];
-  auto sum_dzxh = w5[1];
-  auto k1 = w5[2];
-  auto k2 = w5[3];
-  auto k3 = w5[4];
+      : torch::empty({2 * BN_GM_MAX + 3, (long)g.C}, fopt);
+  TORCH_CHECK(w5.size(0) >= 2 * BN_GM_MAX + 3 && w5.size(1) == g.C &&
+              w5.is_contiguous());
+  float* wp = w5.data_ptr<float>();
+  float* partial_dz = wp;
+  float* partial_dzxh = wp + (long)BN_GM_MAX * g.C;
+  auto k1 = w5[2 * BN_GM_MAX + 0];
+  auto k2 = w5[2 * BN_GM_MAX + 1];
+  auto k3 = w5[2 * BN_GM_MAX + 2];
   // fresh allocations: returned to autograd as parameter gradients
   auto dweight = torch::empty({g.C}, fopt);
   auto dbias = torch::empty({g.C}, fopt);
@@ -143,7 +152,8 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   torch::Tensor dres;
   if (need_dres) dres = torch::empty_like(x);
   dim3 block(BLOCK_THREADS);
-  dim3 grid_r(bn_grid_m(g, 1024), g.grid_c);
+  int gm = bn_grid_m(g, BN_GM_MAX);
+  dim3 grid_r(gm, g.grid_c);
   dim3 grid_a(bn_grid_m(g, 4096), g.grid_c);
   auto st = cur_stream();
 
@@ -154,18 +164,16 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                          st, GET(x), GET(dy), GET(y),                         \
                          save_mean.data_ptr<float>(),                         \
                          save_rstd.data_ptr<float>(),                         \
-                         sum_dz.data_ptr<float>(),                            \
-                         sum_dzxh.data_ptr<float>(), g.M, g.C);               \
+                         partial_dz, partial_dzxh, g.M, g.C);                 \
     else                                                                      \
       hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid_r, block, 0,  \
                          st, GET(x), GET(dy), GET(y),                         \
                          save_mean.data_ptr<float>(),                         \
                          save_rstd.data_ptr<float>(),                         \
-                         sum_dz.data_ptr<float>(),                            \
-                         sum_dzxh.data_ptr<float>(), g.M, g.C);               \
+                         partial_dz, partial_dzxh, g.M, g.C);                 \
     hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((g.C + 255) / 256),       \
-                       dim3(256), 0, st, sum_dz.data_ptr<float>(),            \
-                       sum_dzxh.data_ptr<float>(), weight.data_ptr<float>(),  \
+                       dim3(256), 0, st, partial_dz, partial_dzxh, gm,        \
+                       weight.data_ptr<float>(),                              \
                        save_rstd.data_ptr<float>(), k1.data_ptr<float>(),     \
                        k2.data_ptr<float>(), k3.data_ptr<float>(),            \
                        dweight.data_ptr<float>(), dbias.data_ptr<float>(),    \

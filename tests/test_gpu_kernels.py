@@ -154,20 +154,37 @@ def test_fused_bn_gpu_vs_cpu_reference(ext):
                        relu=True, residual=rg)
     y.backward(gy.to(dev, torch.bfloat16).contiguous(
         memory_format=torch.channels_last))
-    tol = 5e-2  # bf16 inputs
+    tol = 5e-2  # bf16 inputs vs fp32 reference
     assert torch.allclose(y.detach().float().cpu(), y_ref.detach(), atol=tol)
     assert torch.allclose(rm_g.cpu(), rm_c, atol=1e-2)
     assert torch.allclose(rv_g.cpu(), rv_c, atol=1e-2)
-    # dx/dres: near the ReLU boundary (|y|~bf16 eps) the fp32 and bf16 masks
-    # legitimately disagree; compare away from the boundary.
-    interior = z_ref.detach().abs() > 1e-2
-    dxe = (xg.grad.float().cpu() - xc.grad).abs()
-    assert dxe[interior].max() < tol, dxe[interior].max()
-    dre = (rg.grad.float().cpu() - rc.grad).abs()
-    assert dre[interior].max() < tol, dre[interior].max()
-    assert torch.allclose(wg.grad.cpu(), wc.grad, atol=0.5), \
-        (wg.grad.cpu() - wc.grad).abs().max()
-    assert torch.allclose(bg.grad.cpu(), bc.grad, atol=0.5)
+    # Backward: compare against a torch fp32 reference computed from the
+    # KERNEL'S saved forward tensors (same mask / same quantized inputs), so
+    # relu-boundary bf16 mask flips don't poison summed statistics.
+    xq = xg.detach().float().cpu()
+    yq = y.detach().float().cpu()
+    dyq = gy.to(torch.bfloat16).float()
+    mean_q = torch.zeros(C)
+    rstd_q = torch.zeros(C)
+    # recompute stats exactly like the kernel (fp32 over quantized x)
+    mean_q = xq.mean((0, 2, 3))
+    var_q = xq.var((0, 2, 3), unbiased=False)
+    rstd_q = (var_q + 1e-5).rsqrt()
+    dz = dyq * (yq > 0).float()
+    M = N * H * W
+    xhat = (xq - mean_q[None, :, None, None]) * rstd_q[None, :, None, None]
+    sum_dz = dz.sum((0, 2, 3))
+    sum_dzxh = (dz * xhat).sum((0, 2, 3))
+    dx_ref = (w * rstd_q)[None, :, None, None] * (
+        dz - sum_dz[None, :, None, None] / M
+        - xhat * sum_dzxh[None, :, None, None] / M)
+    assert torch.allclose(xg.grad.float().cpu(), dx_ref, atol=tol), \
+        (xg.grad.float().cpu() - dx_ref).abs().max()
+    assert torch.allclose(rg.grad.float().cpu(), dz.to(torch.bfloat16).float(),
+                          atol=tol)
+    assert torch.allclose(wg.grad.cpu(), sum_dzxh, atol=0.3), \
+        (wg.grad.cpu() - sum_dzxh).abs().max()
+    assert torch.allclose(bg.grad.cpu(), sum_dz, atol=0.3)
 
 
 def test_fused_bn_gpu_fp32_exact(ext):
