@@ -92,6 +92,10 @@ __global__ void bn_fwd_reduce_kernel(const T* __restrict__ x,
 
 // ------------------------------------------------------------ fwd finalize
 // scale = w*rstd; shift = b - mean*scale; running stats updated in place.
+// block = 64 channels x 4 row-lanes: lanes split the grid_m partial rows so
+// small-C layers still read the partial matrix with thousands of threads.
+#define FIN_CH 64
+#define FIN_LANES 4
 __global__ void bn_fwd_finalize_kernel(const float* __restrict__ partial_sum,
                                        const float* __restrict__ partial_sq,
                                        int grid_m,
@@ -104,12 +108,24 @@ __global__ void bn_fwd_finalize_kernel(const float* __restrict__ partial_sum,
                                        float* __restrict__ scale,
                                        float* __restrict__ shift, long M,
                                        int C, float eps, float momentum) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int tx = threadIdx.x % FIN_CH;
+  int ty = threadIdx.x / FIN_CH;
+  int c = blockIdx.x * FIN_CH + tx;
   if (c >= C) return;
   float s = 0.f, q = 0.f;
-  for (int r = 0; r < grid_m; ++r) {
+  for (int r = ty; r < grid_m; r += FIN_LANES) {
     s += partial_sum[(long)r * C + c];
     q += partial_sq[(long)r * C + c];
+  }
+  __shared__ float ls[FIN_LANES * FIN_CH], lq[FIN_LANES * FIN_CH];
+  ls[ty * FIN_CH + tx] = s;
+  lq[ty * FIN_CH + tx] = q;
+  __syncthreads();
+  if (ty != 0) return;
+#pragma unroll
+  for (int r = 1; r < FIN_LANES; ++r) {
+    s += ls[r * FIN_CH + tx];
+    q += lq[r * FIN_CH + tx];
   }
   float mean = s / (float)M;
   float var = fmaxf(q / (float)M - mean * mean, 0.f);
@@ -234,12 +250,24 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_dz,
                                        float* __restrict__ dweight,
                                        float* __restrict__ dbias, long M,
                                        int C) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int tx = threadIdx.x % FIN_CH;
+  int ty = threadIdx.x / FIN_CH;
+  int c = blockIdx.x * FIN_CH + tx;
   if (c >= C) return;
   float sdz = 0.f, sdzxh = 0.f;
-  for (int r = 0; r < grid_m; ++r) {
+  for (int r = ty; r < grid_m; r += FIN_LANES) {
     sdz += partial_dz[(long)r * C + c];
     sdzxh += partial_dzxh[(long)r * C + c];
+  }
+  __shared__ float ls[FIN_LANES * FIN_CH], lq[FIN_LANES * FIN_CH];
+  ls[ty * FIN_CH + tx] = sdz;
+  lq[ty * FIN_CH + tx] = sdzxh;
+  __syncthreads();
+  if (ty != 0) return;
+#pragma unroll
+  for (int r = 1; r < FIN_LANES; ++r) {
+    sdz += ls[r * FIN_CH + tx];
+    sdzxh += lq[r * FIN_CH + tx];
   }
   k1[c] = weight[c] * save_rstd[c];
   k2[c] = sdz / (float)M;

@@ -45,6 +45,17 @@ inline int bn_grid_m(const BNGeom& g, long target_blocks) {
 // max partial rows for the two-stage BN reductions (2 blocks per CU)
 #define BN_GM_MAX 512
 
+// reduce-stage grid: >=32 row-iterations per block so the partial matrix
+// (finalize traffic) stays small relative to the tensor itself
+inline int bn_reduce_gm(const BNGeom& g) {
+  long rows = (g.M + g.rows_per_blk - 1) / g.rows_per_blk;
+  long gm = (rows + 31) / 32;
+  if (gm < 64) gm = 64;
+  if (gm > BN_GM_MAX) gm = BN_GM_MAX;
+  if (gm > rows) gm = rows;
+  return (int)gm;
+}
+
 std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         torch::Tensor bias,
                                         torch::Tensor running_mean,
@@ -72,7 +83,7 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
   auto shift = w6[2 * BN_GM_MAX + 3];
   auto y = torch::empty_like(x);
   dim3 block(BLOCK_THREADS);
-  int gm = bn_grid_m(g, BN_GM_MAX);
+  int gm = bn_reduce_gm(g);
   dim3 grid_r(gm, g.grid_c);
   dim3 grid_a(bn_grid_m(g, 4096), g.grid_c);
   auto st = cur_stream();
@@ -81,8 +92,10 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
   {                                                                           \
     hipLaunchKernelGGL((bn_fwd_reduce_kernel<T>), grid_r, block, 0, st,       \
                        GET(x), partial_sum, partial_sq, g.M, g.C);            \
-    hipLaunchKernelGGL(bn_fwd_finalize_kernel, dim3((g.C + 255) / 256),       \
-                       dim3(256), 0, st, partial_sum, partial_sq, gm,         \
+    hipLaunchKernelGGL(bn_fwd_finalize_kernel,                                \
+                       dim3((g.C + FIN_CH - 1) / FIN_CH),                     \
+                       dim3(FIN_CH * FIN_LANES), 0, st, partial_sum,          \
+                       partial_sq, gm,                                        \
                        weight.data_ptr<float>(),                              \
                        bias.data_ptr<float>(), running_mean.data_ptr<float>(),\
                        running_var.data_ptr<float>(),                         \
@@ -153,7 +166,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   torch::Tensor dres;
   if (need_dres) dres = torch::empty_like(x);
   dim3 block(BLOCK_THREADS);
-  int gm = bn_grid_m(g, BN_GM_MAX);
+  int gm = bn_reduce_gm(g);
   dim3 grid_r(gm, g.grid_c);
   dim3 grid_a(bn_grid_m(g, 4096), g.grid_c);
   auto st = cur_stream();
@@ -172,8 +185,10 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                          save_mean.data_ptr<float>(),                         \
                          save_rstd.data_ptr<float>(),                         \
                          partial_dz, partial_dzxh, g.M, g.C);                 \
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((g.C + 255) / 256),       \
-                       dim3(256), 0, st, partial_dz, partial_dzxh, gm,        \
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel,                                \
+                       dim3((g.C + FIN_CH - 1) / FIN_CH),                     \
+                       dim3(FIN_CH * FIN_LANES), 0, st, partial_dz,           \
+                       partial_dzxh, gm,                                      \
                        weight.data_ptr<float>(),                              \
                        save_rstd.data_ptr<float>(), k1.data_ptr<float>(),     \
                        k2.data_ptr<float>(), k3.data_ptr<float>(),            \
