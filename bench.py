@@ -24,8 +24,8 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch-size", type=int, default=256,
-                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--batch-size", type=int, default=512,
+                   help="per-GPU batch (weak scaling; 288 GB HBM3E/GPU)")
     p.add_argument("--model", default="resnet50")
     p.add_argument("--strategy", default="AllReduce")
     p.add_argument("--bucket-mb", type=int, default=25)
