@@ -220,6 +220,7 @@ class DistributedEngine:
         self._sync_initial_params()
         self._build_buckets_and_hooks()
         self._patch_optimizer()
+        self.graph_item._engine = self  # checkpoint Saver discovery
         self._setup_done = True
         return self
 
@@ -584,6 +585,179 @@ class DistributedEngine:
             cls = "SparseAdam"
         apply_mod.apply_sparse_rows(cls, plan.param.data, indices, values,
                                     sh.state, plan.hyper)
+
+    # ----------------------------------------------------- checkpoint support
+    def _state_keys(self, cls_name: str, hyper: dict):
+        """Deterministic optimizer-state key list for one var/shard."""
+        if cls_name == "SGD":
+            return ["momentum_buffer"] if hyper.get("momentum", 0) else []
+        if cls_name in ("Adam", "AdamW"):
+            keys = ["step", "exp_avg", "exp_avg_sq"]
+            if hyper.get("amsgrad"):
+                keys.append("max_exp_avg_sq")
+            return keys
+        if cls_name == "Adagrad":
+            return ["step", "sum"]
+        if cls_name == "RMSprop":
+            keys = ["square_avg"]
+            if hyper.get("momentum", 0) > 0:
+                keys.append("momentum_buffer")
+            if hyper.get("centered"):
+                keys.append("grad_avg")
+            return keys
+        return []
+
+    def consolidated_optimizer_state(self) -> Dict[str, dict]:
+        """Full (unsharded) optimizer state per variable name, assembled on
+        EVERY rank (PS shard states are broadcast from their owners).
+
+        This is what makes checkpoints single-node compatible — the
+        reference's SaveSliceInfo reassembly (partitioner.py:292-346).
+        """
+        out: Dict[str, dict] = {}
+        param_to_bucket = {}
+        for b in self.buckets:
+            for p, off in zip(b.params, b.offsets):
+                param_to_bucket[id(p)] = (b, off)
+        for plan in self.var_plans:
+            keys = self._state_keys(plan.cls_name, plan.hyper)
+            if not keys:
+                out[plan.name] = {}
+                continue
+            state: dict = {}
+            if plan.bucketed:
+                b, off = param_to_bucket[id(plan.param)]
+                n = plan.param.numel()
+                for k in keys:
+                    if k not in b.state:
+                        continue
+                    t = b.state[k]
+                    if t.dim() == 0:  # shared step counter
+                        state[k] = t.clone()
+                    else:
+                        state[k] = t[off:off + n].view(plan.param.shape).clone()
+            elif plan.sparse or (len(plan.shards) == 1
+                                 and plan.shards[0].kind == "allreduce"
+                                 and plan.shards[0].slice is None):
+                sh = plan.shards[0]
+                if sh.state:
+                    for k in keys:
+                        if k in sh.state:
+                            state[k] = sh.state[k].clone()
+            else:
+                # sharded (partitioned AR or PS): assemble along the axis
+                pieces: Dict[str, list] = {k: [] for k in keys}
+                step_val = None
+                for sh in plan.shards:
+                    shard_shape = (sh.slice.view(plan.param.data).shape
+                                   if sh.slice else plan.param.shape)
+                    if sh.kind == "allreduce":
+                        src_state = sh.state or {}
+                        for k in keys:
+                            if k == "step":
+                                if k in src_state:
+                                    step_val = src_state[k].clone()
+                                continue
+                            if k in src_state:
+                                pieces[k].append(src_state[k])
+                    else:  # PS shard: broadcast from owner
+                        present = torch.zeros(len(keys), dtype=torch.uint8,
+                                              device=self.device)
+                        if self.rank == sh.owner_rank and sh.state:
+                            for i, k in enumerate(keys):
+                                present[i] = int(k in sh.state)
+                        if self.world_size > 1:
+                            dist.broadcast(present, src=sh.owner_rank,
+                                           group=self.process_group)
+                        for i, k in enumerate(keys):
+                            if not bool(present[i]):
+                                continue
+                            if k == "step":
+                                buf = torch.zeros((), device=self.device)
+                                if self.rank == sh.owner_rank:
+                                    buf.copy_(sh.state[k])
+                                if self.world_size > 1:
+                                    dist.broadcast(buf, src=sh.owner_rank,
+                                                   group=self.process_group)
+                                step_val = buf.cpu()
+                                continue
+                            buf = torch.zeros(shard_shape, device=self.device)
+                            if self.rank == sh.owner_rank:
+                                buf.copy_(sh.state[k])
+                            if self.world_size > 1:
+                                dist.broadcast(buf, src=sh.owner_rank,
+                                               group=self.process_group)
+                            pieces[k].append(buf)
+                axis = plan.shards[0].slice.axis if plan.shards[0].slice else 0
+                for k in keys:
+                    if k == "step":
+                        if step_val is not None:
+                            state[k] = step_val
+                    elif pieces[k]:
+                        state[k] = torch.cat(
+                            [t.to(self.device) for t in pieces[k]],
+                            dim=axis).clone()
+            out[plan.name] = state
+        return out
+
+    def load_optimizer_state(self, full_state: Dict[str, dict]):
+        """Distribute a full optimizer-state dict back into bucket flats and
+        shard owners (inverse of consolidated_optimizer_state)."""
+        param_to_bucket = {}
+        for b in self.buckets:
+            for p, off in zip(b.params, b.offsets):
+                param_to_bucket[id(p)] = (b, off)
+        for plan in self.var_plans:
+            state = full_state.get(plan.name) or {}
+            if not state:
+                continue
+            keys = self._state_keys(plan.cls_name, plan.hyper)
+            if plan.bucketed:
+                b, off = param_to_bucket[id(plan.param)]
+                n = plan.param.numel()
+                if not b.state:
+                    b.state = apply_mod.make_state(b.cls_name, b.flat_param,
+                                                   b.hyper)
+                for k in keys:
+                    if k not in state:
+                        continue
+                    src = state[k]
+                    if k == "step":
+                        b.state[k] = src.to(b.state.get(k, src).dtype) \
+                            if "step" in b.state else src.clone()
+                    else:
+                        if k not in b.state:
+                            b.state[k] = torch.zeros_like(b.flat)
+                        b.state[k][off:off + n].view(plan.param.shape).copy_(src)
+            else:
+                axis = plan.shards[0].slice.axis if plan.shards[0].slice else 0
+                for sh in plan.shards:
+                    applies_here = (sh.kind == "allreduce"
+                                    or self.rank == sh.owner_rank
+                                    or plan.sparse)
+                    if not applies_here:
+                        continue
+                    ref = sh.master if sh.master is not None else (
+                        sh.slice.view(plan.param.data) if sh.slice
+                        else plan.param.data)
+                    if sh.state is None:
+                        sh.state = apply_mod.make_state(plan.cls_name, ref,
+                                                        plan.hyper)
+                    for k in keys:
+                        if k not in state:
+                            continue
+                        src = state[k]
+                        if k == "step":
+                            sh.state[k] = src.clone()
+                        else:
+                            piece = (sh.slice.view(src.to(self.device))
+                                     if sh.slice else src.to(self.device))
+                            sh.state[k] = piece.clone().contiguous()
+                    if sh.master is not None:
+                        view = sh.slice.view(plan.param.data) if sh.slice \
+                            else plan.param.data
+                        sh.master.copy_(view)
+                        sh.stage.copy_(view)
 
     # ------------------------------------------------------------------ misc
     @property
