@@ -256,12 +256,14 @@ class DistributedEngine:
             cls_name, hyper = self._hyper_for(item.param)
             plan = VarPlan(name=node.var_name, param=item.param,
                            sparse=item.is_sparse, cls_name=cls_name, hyper=hyper)
-            if node.part_config:
+            if node.part_config and not getattr(
+                    item.param, "_autodist_shard_local", False):
                 slices = make_shard_slices(item.shape, node.partitioner)
                 for sl, part in zip(slices, node.part_config):
                     plan.shards.append(self._make_shard(part, sl))
             else:
-                plan.shards.append(self._make_shard(node, None))
+                plan.shards.append(self._make_shard(node, None,
+                                                    param=item.param))
             self.var_plans.append(plan)
         known = {p.name for p in self.var_plans}
         missing = [n for n in vars_by_name if n not in known
@@ -284,7 +286,12 @@ class DistributedEngine:
             logging.info("unsupported optimizer %s: falling back to user "
                          "optimizer.step() after gradient sync", unsupported)
 
-    def _make_shard(self, node, sl: Optional[ShardSlice]) -> ShardPlan:
+    def _make_shard(self, node, sl: Optional[ShardSlice],
+                    param=None) -> ShardPlan:
+        if param is not None and getattr(param, "_autodist_shard_local", False):
+            # exclusively-owned shard (e.g. ShardedEmbedding): update locally,
+            # never synchronize
+            return ShardPlan(name=node.var_name, kind="local", slice=None)
         if node.all_reduce_synchronizer is not None:
             s = node.all_reduce_synchronizer
             return ShardPlan(name=node.var_name, kind="allreduce", slice=sl,
@@ -302,6 +309,8 @@ class DistributedEngine:
         if self.world_size <= 1:
             return
         for plan in self.var_plans:
+            if getattr(plan.param, "_autodist_shard_local", False):
+                continue  # exclusively-owned shard: shapes differ per rank
             dist.broadcast(plan.param.data, src=0, group=self.process_group)
 
     def _build_buckets_and_hooks(self):
@@ -313,6 +322,10 @@ class DistributedEngine:
                 continue
             whole = len(plan.shards) == 1 and plan.shards[0].slice is None
             sh0 = plan.shards[0]
+            if sh0.kind == "local":
+                sh0.state = apply_mod.make_state(plan.cls_name,
+                                                 plan.param.data, plan.hyper)
+                continue
             if whole and sh0.kind == "allreduce":
                 plan.bucketed = True
                 bucket_items.append((plan.param, sh0.group, sh0.compressor,
@@ -463,7 +476,9 @@ class DistributedEngine:
             if plan.sparse or plan.bucketed:
                 continue
             for sh in plan.shards:
-                if sh.kind != "allreduce":
+                if sh.kind not in ("allreduce", "local"):
+                    continue
+                if sh.kind == "local" and plan.param.grad is None:
                     continue
                 if sh.slice is None:
                     p, g = plan.param.data, plan.param.grad
@@ -637,7 +652,8 @@ class DistributedEngine:
                     else:
                         state[k] = t[off:off + n].view(plan.param.shape).clone()
             elif plan.sparse or (len(plan.shards) == 1
-                                 and plan.shards[0].kind == "allreduce"
+                                 and plan.shards[0].kind in ("allreduce",
+                                                             "local")
                                  and plan.shards[0].slice is None):
                 sh = plan.shards[0]
                 if sh.state:
