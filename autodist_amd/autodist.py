@@ -151,10 +151,31 @@ class AutoDist:
 
     def _setup(self, strategy: Strategy, world: int):
         """Launch workers if this chief owns the launch (reference _setup,
-        autodist.py:120-128)."""
+        autodist.py:120-128). Single node: local subprocesses via the
+        Coordinator; multi node: SSH launch via the Cluster with the strategy
+        file shipped to every node (reference coordinator.py:84-88)."""
         if "RANK" in os.environ or world <= 1:
             return
         if not is_chief():
+            return
+        if self._resource_spec.num_nodes > 1:
+            import sys
+            from autodist_amd.runtime.cluster import SSHCluster
+            cluster = SSHCluster(self._resource_spec)
+            os.environ.update({
+                "RANK": "0", "LOCAL_RANK": "0",
+                "WORLD_SIZE": str(cluster.world_size),
+                "MASTER_ADDR": cluster.master_addr,
+                "MASTER_PORT": str(cluster.master_port),
+            })
+            strategy_path = strategy.serialize()
+            for addr in self._resource_spec.nodes:
+                if addr != self._resource_spec.chief:
+                    cluster.remote_copy(addr, strategy_path,
+                                        os.path.dirname(strategy_path))
+            cluster.start([sys.executable] + sys.argv,
+                          extra_env={"AUTODIST_STRATEGY_ID": strategy.id})
+            self._coordinator = cluster
             return
         port = find_free_port()
         os.environ.update({
@@ -166,12 +187,12 @@ class AutoDist:
         self._coordinator.launch_clients(world, port)
 
     def _build(self):
+        from autodist_amd.parallel.graph_transformer import GraphTransformer
         strategy = self._build_or_load_strategy()
         world = self._decide_world()
         self._setup(strategy, world)
         compiled = self._compile_strategy(strategy)
-        self._engine = DistributedEngine(self.graph_item, compiled)
-        self._engine.setup()
+        self._engine = GraphTransformer(compiled, self.graph_item).transform()
         remapper = Remapper(self._engine.rank, self._engine.world_size,
                             self._engine.device, self._engine.process_group)
         self._session = WrappedSession(self._engine, remapper, self.graph_item)
