@@ -40,9 +40,6 @@ from autodist_amd.parallel.partitioner import ShardSlice, make_shard_slices
 from autodist_amd.proto.strategy_ir import CompressorType
 from autodist_amd.utils import logging
 
-_ASYNC_PS_MAX_DEPTH = 4  # queue cap for sync=False (unbounded staleness)
-
-
 @dataclasses.dataclass
 class ShardPlan:
     name: str
@@ -143,15 +140,6 @@ class ShardReducer:
                 self.plan.slice.view(self.param.grad).copy_(self._reduced_tensor)
 
 
-class _PSRound:
-    """One in-flight PS reduce->apply->broadcast round (staleness pipeline)."""
-
-    def __init__(self, shard: ShardPlan, event=None, handle=None):
-        self.shard = shard
-        self.event = event     # GPU: comm-stream event after broadcast
-        self.handle = handle   # CPU: async work handle of broadcast
-
-
 class DistributedEngine:
     """One rank's engine. Builds plans from a compiled Strategy and runs the
     per-step synchronization + update protocol."""
@@ -180,7 +168,7 @@ class DistributedEngine:
         self.buckets: List[Bucket] = []
         self.var_plans: List[VarPlan] = []
         self._hook_handles = []
-        self._ps_outstanding: Dict[str, List[_PSRound]] = {}
+        self._ps_outstanding: Dict[str, list] = {}
         self._replica_ranks: Dict[str, int] = {}
         self._step_count = 0
         self._fallback_user_opt = False
@@ -493,66 +481,17 @@ class DistributedEngine:
             states = [x[2] for x in items]
             apply_mod.apply_dense(cls_name, params, grads, states, items[0][3])
 
-    # -- PS path -----------------------------------------------------------
+    # -- PS path (delegates to parallel/ps_synchronizer.py) -----------------
     def _ps_key(self, sh: ShardPlan) -> str:
         return sh.name + f"/{sh.slice.start if sh.slice else 0}"
 
     def _issue_ps_round(self, plan: VarPlan, sh: ShardPlan):
-        grad = plan.param.grad
-        if grad is None:
-            return
-        gview = sh.slice.view(grad) if sh.slice else grad
-        gbuf = gview if gview.is_contiguous() else gview.contiguous()
-
-        def round_body():
-            gbuf.mul_(1.0 / self.world_size)
-            dist.reduce(gbuf, dst=sh.owner_rank, group=self.process_group)
-            if self.rank == sh.owner_rank:
-                apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
-                                      [sh.state], plan.hyper)
-                src_buf = sh.master
-            else:
-                src_buf = sh.stage
-            dist.broadcast(src_buf, src=sh.owner_rank, group=self.process_group)
-            if self.rank == sh.owner_rank:
-                sh.stage.copy_(sh.master)
-
-        if self.world_size <= 1:
-            # degenerate single-rank PS: apply to master, stage it
-            gbuf2 = gbuf
-            apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf2],
-                                  [sh.state], plan.hyper)
-            sh.stage.copy_(sh.master)
-            self._ps_outstanding[self._ps_key(sh)].append(_PSRound(sh))
-            return
-        if self.device.type == "cuda":
-            ev = torch.cuda.Event()
-            ev.record()
-            self.comm_stream.wait_event(ev)
-            with torch.cuda.stream(self.comm_stream):
-                round_body()
-            done = torch.cuda.Event()
-            done.record(self.comm_stream)
-            self._ps_outstanding[self._ps_key(sh)].append(_PSRound(sh, event=done))
-        else:
-            round_body()
-            self._ps_outstanding[self._ps_key(sh)].append(_PSRound(sh))
+        from autodist_amd.parallel.ps_synchronizer import PSSynchronizer
+        PSSynchronizer.issue_round(self, plan, sh)
 
     def _consume_ps_rounds(self, plan: VarPlan, sh: ShardPlan):
-        """Pop rounds older than the staleness bound and install their
-        results into the live parameter (reference staleness queues,
-        ps_synchronizer.py:388-458; sync barrier 335-385)."""
-        key = self._ps_key(sh)
-        rounds = self._ps_outstanding[key]
-        depth = sh.staleness if sh.sync else _ASYNC_PS_MAX_DEPTH
-        while len(rounds) > depth:
-            r = rounds.pop(0)
-            if r.event is not None:
-                torch.cuda.current_stream().wait_event(r.event)
-            if r.handle is not None:
-                r.handle.wait()
-            view = sh.slice.view(plan.param.data) if sh.slice else plan.param.data
-            view.copy_(sh.stage)
+        from autodist_amd.parallel.ps_synchronizer import PSSynchronizer
+        PSSynchronizer.consume_due_rounds(self, plan, sh)
 
     def drain(self):
         """Consume ALL outstanding PS rounds (end of training / checkpoint)."""
