@@ -12,6 +12,15 @@ Reference API parity: `AutoDist(resource_spec_file, strategy_builder)`,
 """
 __version__ = "0.1.0"
 
+# torch version gate (reference gates TF 1.15-2.2, __init__.py:35-42)
+import torch as _torch
+
+_major, _minor = (int(x) for x in _torch.__version__.split(".")[:2])
+if (_major, _minor) < (2, 1):
+    raise RuntimeError(
+        f"autodist_amd requires torch>=2.1 (post-accumulate-grad hooks); "
+        f"found {_torch.__version__}")
+
 from autodist_amd.autodist import AutoDist, get_default_autodist
 from autodist_amd.const import ENV
 
