@@ -6,6 +6,7 @@
 
 #include "multi_tensor.hip"
 #include "bn_ops.hip"
+#include "psgd_gemm.hip"
 
 namespace {
 
@@ -368,6 +369,61 @@ torch::Tensor gather_rows(torch::Tensor src, torch::Tensor idx) {
   return out;
 }
 
+// ------------------------------------------------------------- PowerSGD
+torch::Tensor psgd_mq(torch::Tensor M2d, torch::Tensor Qp) {
+  long n = M2d.size(0), s = M2d.size(1);
+  TORCH_CHECK(n % 64 == 0 && s % 64 == 0, "psgd_mq needs 64-multiples");
+  TORCH_CHECK(Qp.size(0) == s && Qp.size(1) == PSGD_R);
+  auto C = torch::zeros({n, (long)PSGD_R},
+                        M2d.options().dtype(torch::kFloat32));
+  long gridx = n / 64;
+  long chunks = s / PSGD_BK;
+  long splitk = std::min(std::max<long>((512 + gridx - 1) / gridx, 1), chunks);
+  hipLaunchKernelGGL(psgd_mq_kernel, dim3(gridx, splitk), dim3(256), 0,
+                     cur_stream(), M2d.data_ptr<float>(),
+                     Qp.contiguous().data_ptr<float>(), C.data_ptr<float>(),
+                     n, s);
+  return C;
+}
+
+torch::Tensor psgd_mtp(torch::Tensor M2d, torch::Tensor Pp) {
+  long n = M2d.size(0), s = M2d.size(1);
+  TORCH_CHECK(n % 64 == 0 && s % 64 == 0, "psgd_mtp needs 64-multiples");
+  TORCH_CHECK(Pp.size(0) == n && Pp.size(1) == PSGD_R);
+  auto C = torch::zeros({s, (long)PSGD_R},
+                        M2d.options().dtype(torch::kFloat32));
+  long gridx = s / 64;
+  long chunks = n / PSGD_BK;
+  long splitk = std::min(std::max<long>((512 + gridx - 1) / gridx, 1), chunks);
+  hipLaunchKernelGGL(psgd_mtp_kernel, dim3(gridx, splitk), dim3(256), 0,
+                     cur_stream(), M2d.data_ptr<float>(),
+                     Pp.contiguous().data_ptr<float>(), C.data_ptr<float>(),
+                     n, s);
+  return C;
+}
+
+void psgd_decompress_ef(torch::Tensor flat, torch::Tensor err,
+                        torch::Tensor m_local, torch::Tensor Pp,
+                        torch::Tensor Qp, double scale) {
+  long numel = flat.numel();
+  long s = m_local.size(1);
+  hipLaunchKernelGGL(psgd_decompress_ef_kernel, dim3(grid_for(numel, 1)),
+                     dim3(BLOCK_THREADS), 0, cur_stream(),
+                     flat.data_ptr<float>(), err.data_ptr<float>(),
+                     m_local.data_ptr<float>(), Pp.data_ptr<float>(),
+                     Qp.data_ptr<float>(), numel, s, (float)scale);
+}
+
+void psgd_add_err_pad(torch::Tensor flat, torch::Tensor err,
+                      torch::Tensor out) {
+  long numel = flat.numel();
+  long total = out.numel();
+  hipLaunchKernelGGL(psgd_add_err_pad_kernel, dim3(grid_for(total, 1)),
+                     dim3(BLOCK_THREADS), 0, cur_stream(),
+                     flat.data_ptr<float>(), err.data_ptr<float>(),
+                     out.data_ptr<float>(), numel, total);
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -385,4 +441,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd", &bn_bwd,
         "fused NHWC batchnorm bwd (+relu-mask+dres), returns "
         "dx/dweight/dbias[/dres]");
+  m.def("psgd_mq", &psgd_mq, "PowerSGD P = M @ Q (MFMA f32 16x16x4)");
+  m.def("psgd_mtp", &psgd_mtp, "PowerSGD Qn = M^T @ P (MFMA f32 16x16x4)");
+  m.def("psgd_decompress_ef", &psgd_decompress_ef,
+        "fused hat = P Q^T * scale; err = M - hat; flat = hat");
+  m.def("psgd_add_err_pad", &psgd_add_err_pad,
+        "padded M = flat + err (zero tail)");
 }

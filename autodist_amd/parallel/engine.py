@@ -188,9 +188,13 @@ class DistributedEngine:
             torch.cuda.set_device(self.device)
         addr = os.environ.get("MASTER_ADDR", DEFAULT_MASTER_ADDR)
         port = os.environ.get("MASTER_PORT", str(DEFAULT_MASTER_PORT))
+        # generous timeout: the first collective can sit behind minutes of
+        # MIOpen find-mode tuning on each rank
+        import datetime
         dist.init_process_group(
             backend=backend, init_method=f"tcp://{addr}:{port}",
-            rank=self.rank, world_size=self.world_size)
+            rank=self.rank, world_size=self.world_size,
+            timeout=datetime.timedelta(minutes=45))
         logging.info("process group ready: backend=%s rank=%d world=%d",
                      backend, self.rank, self.world_size)
 

@@ -49,9 +49,16 @@ class PowerSGDCompressor(Compressor):
         self._pad: Optional[torch.Tensor] = None
         self._side = 0
 
+    def _use_hip(self, flat) -> bool:
+        from autodist_amd.ops import api as ops_api
+        return (flat.is_cuda and flat.dtype == torch.float32
+                and self.rank <= 16 and ops_api.has_gpu_ops())
+
     def _init(self, flat: torch.Tensor):
         numel = flat.numel()
-        self._side = int(math.ceil(math.sqrt(numel)))
+        side = int(math.ceil(math.sqrt(numel)))
+        # gfx950 MFMA kernels tile in 64s; rounding up costs <3% padding
+        self._side = (side + 63) // 64 * 64
         self._error = torch.zeros_like(flat)
         self._pad = torch.zeros(self._side * self._side, dtype=flat.dtype,
                                 device=flat.device)
@@ -59,18 +66,34 @@ class PowerSGDCompressor(Compressor):
         g = torch.Generator(device="cpu").manual_seed(seed)
         q = torch.randn(self._side, self.rank, generator=g)
         self._q = _orthonormalize(q.to(flat.device, flat.dtype))
+        if self._use_hip(flat):
+            self._qpad = torch.zeros(self._side, 16, device=flat.device)
+            self._ppad = torch.zeros(self._side, 16, device=flat.device)
 
     def reduce(self, flat, group=None, async_op=False, scale=1.0):
         if self._error is None:
             self._init(flat)
         n = flat.numel()
-        flat.add_(self._error)
-        self._pad[:n].copy_(flat)
         m = self._pad.view(self._side, self._side)
-        p = m @ self._q                                   # [side, r]
+        if self._use_hip(flat):
+            from autodist_amd.ops import api as ops_api
+            ext = ops_api.ext()
+            ext.psgd_add_err_pad(flat, self._error, self._pad)
+            self._qpad[:, :self.rank].copy_(self._q)
+            p = ext.psgd_mq(m, self._qpad)[:, :self.rank].contiguous()
+        else:
+            flat.add_(self._error)
+            self._pad[:n].copy_(flat)
+            p = m @ self._q
         dist.all_reduce(p, op=dist.ReduceOp.SUM, group=group)
         _orthonormalize(p)
-        q = m.t() @ p                                     # [side, r]
+        if self._use_hip(flat):
+            from autodist_amd.ops import api as ops_api
+            ext = ops_api.ext()
+            self._ppad[:, :self.rank].copy_(p)
+            q = ext.psgd_mtp(m, self._ppad)[:, :self.rank].contiguous()
+        else:
+            q = m.t() @ p
         handle = dist.all_reduce(q, op=dist.ReduceOp.SUM, group=group,
                                  async_op=async_op)
         self._p = p
@@ -84,6 +107,15 @@ class PowerSGDCompressor(Compressor):
         if h is not None:
             h.wait()
         n = flat.numel()
+        if self._use_hip(flat):
+            from autodist_amd.ops import api as ops_api
+            self._qpad.zero_()
+            self._qpad[:, :self.rank].copy_(q)
+            m = self._pad.view(self._side, self._side)
+            ops_api.ext().psgd_decompress_ef(flat, self._error, m,
+                                             self._ppad, self._qpad,
+                                             self._scale)
+            return
         hat = (self._p @ q.t()).mul_(self._scale).view(-1)
         self._error.copy_(flat).sub_(hat[:n])
         flat.copy_(hat[:n])

@@ -217,6 +217,72 @@ def test_fused_bn_gpu_fp32_exact(ext):
     assert torch.allclose(rv.cpu(), rvc, atol=1e-4)
 
 
+def test_psgd_mfma_gemms(ext):
+    """PowerSGD MFMA factor GEMMs vs torch.matmul (rocBLAS) reference."""
+    torch.manual_seed(0)
+    n, s, r = 192, 256, 4
+    M = torch.randn(n, s, device="cuda")
+    Qp = torch.zeros(s, 16, device="cuda")
+    Qp[:, :r] = torch.randn(s, r, device="cuda")
+    P = ext.psgd_mq(M, Qp)
+    P_ref = M @ Qp
+    assert torch.allclose(P, P_ref, atol=1e-3), (P - P_ref).abs().max()
+    Pp = torch.zeros(n, 16, device="cuda")
+    Pp[:, :r] = torch.randn(n, r, device="cuda")
+    Q2 = ext.psgd_mtp(M, Pp)
+    Q2_ref = M.t() @ Pp
+    assert torch.allclose(Q2, Q2_ref, atol=1e-3), (Q2 - Q2_ref).abs().max()
+
+
+def test_psgd_decompress_ef(ext):
+    torch.manual_seed(1)
+    side = 128
+    numel = side * side - 37
+    flat = torch.randn(numel, device="cuda")
+    err = torch.zeros(numel, device="cuda")
+    m_local = torch.randn(side, side, device="cuda")
+    Pp = torch.randn(side, 16, device="cuda")
+    Qp = torch.randn(side, 16, device="cuda")
+    scale = 0.5
+    hat_ref = (Pp @ Qp.t()).mul(scale).view(-1)[:numel]
+    err_ref = m_local.view(-1)[:numel] - hat_ref
+    ext.psgd_decompress_ef(flat, err, m_local, Pp, Qp, scale)
+    assert torch.allclose(flat, hat_ref, atol=1e-4)
+    assert torch.allclose(err, err_ref, atol=1e-4)
+
+
+def test_psgd_add_err_pad(ext):
+    flat = torch.randn(100, device="cuda")
+    err = torch.randn(100, device="cuda")
+    out = torch.full((128,), 7.0, device="cuda")
+    ext.psgd_add_err_pad(flat, err, out)
+    assert torch.allclose(out[:100], flat + err)
+    assert (out[100:] == 0).all()
+
+
+def test_powersgd_compressor_gpu_roundtrip(monkeypatch):
+    """End-to-end PowerSGD compressor on GPU (collectives stubbed: world=1)
+    — the low-rank projection must match the CPU torch implementation."""
+    import torch.distributed as dist
+    from autodist_amd.parallel import powersgd as psgd_mod
+    monkeypatch.setattr(dist, "all_reduce",
+                        lambda *a, **k: None)
+    from autodist_amd.parallel.powersgd import PowerSGDCompressor
+    torch.manual_seed(5)
+    numel = 5000
+    flat_gpu = torch.randn(numel, device="cuda")
+    flat_cpu = flat_gpu.cpu().clone()
+    cg = PowerSGDCompressor("v", rank=4)
+    cc = PowerSGDCompressor("v", rank=4)
+    for _ in range(3):
+        hg = cg.reduce(flat_gpu, group=None, async_op=False, scale=1.0)
+        cg.finalize(flat_gpu, hg)
+        hc = cc.reduce(flat_cpu, group=None, async_op=False, scale=1.0)
+        cc.finalize(flat_cpu, hc)
+    assert torch.allclose(flat_gpu.cpu(), flat_cpu, atol=1e-2), \
+        (flat_gpu.cpu() - flat_cpu).abs().max()
+
+
 def test_apply_flat_dispatch_uses_hip():
     """apply_flat on GPU must route through the HIP kernel and match the
     CPU torch reference."""
