@@ -506,13 +506,20 @@ class DistributedEngine:
                 if sh.kind == "ps":
                     key = self._ps_key(sh)
                     rounds = self._ps_outstanding.get(key, [])
+                    had_rounds = bool(rounds)
                     while rounds:
                         r = rounds.pop(0)
                         if r.event is not None:
                             torch.cuda.current_stream().wait_event(r.event)
-                    view = sh.slice.view(plan.param.data) if sh.slice \
-                        else plan.param.data
-                    view.copy_(sh.stage)
+                        if r.handle is not None:
+                            for h in (r.handle if isinstance(r.handle, tuple)
+                                      else (r.handle,)):
+                                if h is not None:
+                                    h.wait()
+                    if had_rounds:
+                        view = sh.slice.view(plan.param.data) if sh.slice \
+                            else plan.param.data
+                        view.copy_(sh.stage)
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 

@@ -84,8 +84,24 @@ class PSSynchronizer:
             done.record(engine.comm_stream)
             engine._ps_outstanding[key].append(PSRound(sh, event=done))
         else:
-            round_body()
-            engine._ps_outstanding[key].append(PSRound(sh))
+            # CPU/gloo: async handles so non-owner workers RUN AHEAD within
+            # the staleness bound (the c9-verified behavior) — only the owner
+            # blocks, on the reduced gradient it must apply.
+            gbuf.mul_(1.0 / engine.world_size)
+            h_red = dist.reduce(gbuf, dst=sh.owner_rank,
+                                group=engine.process_group, async_op=True)
+            if engine.rank == sh.owner_rank:
+                h_red.wait()
+                apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
+                                      [sh.state], plan.hyper)
+                sh.stage.copy_(sh.master)
+                src_buf = sh.master
+            else:
+                src_buf = sh.stage
+            h_bc = dist.broadcast(src_buf, src=sh.owner_rank,
+                                  group=engine.process_group, async_op=True)
+            engine._ps_outstanding[key].append(
+                PSRound(sh, handle=(h_red, h_bc)))
 
     @staticmethod
     def consume_due_rounds(engine, plan, sh):
@@ -100,7 +116,10 @@ class PSSynchronizer:
             if r.event is not None:
                 torch.cuda.current_stream().wait_event(r.event)
             if r.handle is not None:
-                r.handle.wait()
+                for h in (r.handle if isinstance(r.handle, tuple)
+                          else (r.handle,)):
+                    if h is not None:
+                        h.wait()
             view = sh.slice.view(plan.param.data) if sh.slice \
                 else plan.param.data
             view.copy_(sh.stage)

@@ -15,7 +15,7 @@ from autodist_amd.utils import logging
 
 def log_graph(name: str, content: str) -> str:
     os.makedirs(DEFAULT_GRAPH_DUMP_DIR, exist_ok=True)
-    ts = datetime.now().strftime("%Y%m%d-%H%M%S")
+    ts = datetime.now().strftime("%Y%m%d-%H%M%S-%f")
     path = os.path.join(DEFAULT_GRAPH_DUMP_DIR, f"{ts}-{name}.txt")
     with open(path, "w", encoding="utf-8") as f:
         f.write(content)

@@ -27,7 +27,7 @@ def _worker(rank, world_size, init_file, fn, args, err_queue):
         raise
 
 
-def run_distributed(fn, world_size=2, args=(), timeout=180):
+def run_distributed(fn, world_size=2, args=(), timeout=300):
     """Run fn(rank, world_size, *args) in `world_size` spawned processes."""
     with tempfile.TemporaryDirectory() as td:
         init_file = os.path.join(td, "rendezvous")

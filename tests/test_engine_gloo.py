@@ -275,6 +275,54 @@ def test_ps_staleness():
     run_distributed(_staleness_case, world_size=2)
 
 
+def _staleness_timing_case(rank, world, staleness):
+    """c9-style timing verification (reference cases/c9.py:92-125): with a
+    deliberately slowed owner rank, a fast non-owner worker's early steps
+    must not block when staleness > 0, and must block when staleness == 0."""
+    import time
+    from autodist_amd.graph_item import GraphItem
+    from autodist_amd.parallel.engine import DistributedEngine
+    from autodist_amd.resource_spec import ResourceSpec
+    from autodist_amd.strategy import PS
+
+    sleep_s = 1.0
+    torch.manual_seed(2)
+    model = torch.nn.Linear(8, 4)
+    g = GraphItem()
+    g.extend_model(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    g.extend_optimizer_info(opt)
+    strategy = PS(staleness=staleness).build(g, ResourceSpec())
+    strategy.graph_config.replicas = [f"127.0.0.1:CPU:{r}" for r in range(world)]
+    engine = DistributedEngine(g, strategy, rank=rank, world_size=world,
+                               device=torch.device("cpu")).setup()
+    # owner of every var is rank 0 (single PS): rank 0 is the SLOW worker
+    t0 = time.perf_counter()
+    for s in range(2):
+        if rank == 0:
+            time.sleep(sleep_s)
+        torch.manual_seed(40 + s)
+        x, y = torch.randn(4, 8), torch.randn(4, 4)
+        opt.zero_grad()
+        torch.nn.functional.mse_loss(model(x), y).backward()
+        opt.step()
+    elapsed = time.perf_counter() - t0
+    engine.drain()
+    if rank == 1:
+        if staleness >= 2:
+            # fast worker ran 2 steps ahead without waiting for the slow owner
+            assert elapsed < sleep_s, f"staleness={staleness}: {elapsed:.2f}s"
+        else:
+            # sync PS: every step waits the slow owner's round
+            assert elapsed > sleep_s, f"staleness={staleness}: {elapsed:.2f}s"
+    engine.teardown()
+
+
+@pytest.mark.parametrize("staleness", [0, 2])
+def test_ps_staleness_timing(staleness):
+    run_distributed(_staleness_timing_case, world_size=2, args=(staleness,))
+
+
 def _feed_fetch_case(rank, world):
     from autodist_amd.remapper import Remapper
     r = Remapper(rank, world, torch.device("cpu"))
