@@ -1,0 +1,81 @@
+"""Small unit tests: collective keys, ENV flags, cost model, failure
+detection (reference §5.2/§5.3: determinism by construction, fail-fast
+coordinator)."""
+import os
+import subprocess
+import sys
+import textwrap
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_collective_keys_deterministic():
+    from autodist_amd.parallel.collective_key import CollectiveKey
+    k1 = CollectiveKey()
+    k2 = CollectiveKey()
+    # instance keys are content-addressed: identical across processes/ranks
+    assert k1.generate_instance_key("layer1.weight") == \
+        k2.generate_instance_key("layer1.weight")
+    assert k1.generate_instance_key("a") != k1.generate_instance_key("b")
+    # group keys increment per distinct device set
+    g1 = k1.generate_group_key(["gpu0", "gpu1"])
+    g2 = k1.generate_group_key(["gpu1", "gpu0"])  # order-insensitive
+    assert g1 == g2
+    assert k1.generate_group_key(["gpu0"]) != g1
+
+
+def test_env_flags(monkeypatch):
+    from autodist_amd.const import ENV, is_chief
+    monkeypatch.delenv("AUTODIST_WORKER", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    assert is_chief()
+    monkeypatch.setenv("AUTODIST_WORKER", "10.0.0.2")
+    assert not is_chief()
+    assert ENV.AUTODIST_WORKER.val == "10.0.0.2"
+    monkeypatch.delenv("AUTODIST_WORKER", raising=False)
+    monkeypatch.setenv("RANK", "3")
+    assert not is_chief()
+    monkeypatch.setenv("AUTODIST_IS_TESTING", "True")
+    assert ENV.AUTODIST_IS_TESTING.val is True
+
+
+def test_cost_model_fit():
+    from autodist_amd.simulator.cost_model import CostModel
+    cm = CostModel()
+    t1 = cm.allreduce_time(100e6, 8)
+    # feed synthetic measurements implying lower efficiency
+    samples = [(100e6, 8, cm.allreduce_time(100e6, 8) * 2)]
+    cm.fit(samples)
+    t2 = cm.allreduce_time(100e6, 8)
+    assert t2 > t1
+    # ps model: sharding owners speeds up the round trip
+    assert cm.ps_round_trip_time(100e6, 8, owners=8) < \
+        cm.ps_round_trip_time(100e6, 8, owners=1)
+
+
+@pytest.mark.integration
+def test_coordinator_fail_fast(tmp_path):
+    """A crashing worker must bring down the chief (reference
+    coordinator.py:98-110: non-zero exit -> os._exit(1))."""
+    script = tmp_path / "crash.py"
+    script.write_text(textwrap.dedent(f"""
+        import os, sys, time
+        sys.path.insert(0, {REPO!r})
+        import torch
+        from autodist_amd import AutoDist
+        from autodist_amd.strategy import AllReduce
+        if os.environ.get("AUTODIST_WORKER"):
+            sys.exit(3)   # worker dies before joining the process group
+        ad = AutoDist(strategy_builder=AllReduce(), world_size=2)
+        with ad.scope():
+            model = torch.nn.Linear(2, 1)
+            opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        sess = ad.create_distributed_session()  # blocks in rendezvous
+        print("SHOULD NOT REACH")
+    """))
+    r = subprocess.run([sys.executable, str(script)], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode != 0
+    assert "SHOULD NOT REACH" not in r.stdout
