@@ -65,7 +65,13 @@ def main():
         if use_cuda else torch.device("cpu")
     if use_cuda:
         torch.cuda.set_device(device)
-        torch.backends.cudnn.benchmark = True  # MIOpen find-best
+        # MIOpen find-best (exhaustive; ~6 min once per conv config).
+        # AUTODIST_FAST_FIND=1 skips it (~2% slower steady state) for quick
+        # runs/profiling.
+        fast_find = os.environ.get("AUTODIST_FAST_FIND") == "1"
+        torch.backends.cudnn.benchmark = not fast_find
+        if fast_find:
+            os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
         torch.backends.cuda.matmul.allow_tf32 = False
 
     from autodist_amd import strategy as strat
