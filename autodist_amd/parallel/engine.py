@@ -104,6 +104,10 @@ class ShardReducer:
     def issue(self, engine):
         if self._issued:
             return
+        # no gradient this step (unused param — consistently unused on every
+        # rank since all ranks run the same graph): skip the collective
+        if self.param.grad is None:
+            return
         self._issued = True
         if engine.world_size <= 1:
             return
@@ -470,7 +474,7 @@ class DistributedEngine:
             for sh in plan.shards:
                 if sh.kind not in ("allreduce", "local"):
                     continue
-                if sh.kind == "local" and plan.param.grad is None:
+                if plan.param.grad is None:  # unused param this step
                     continue
                 if sh.slice is None:
                     p, g = plan.param.data, plan.param.grad
