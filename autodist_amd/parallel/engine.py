@@ -177,6 +177,7 @@ class DistributedEngine:
         self._step_count = 0
         self._fallback_user_opt = False
         self._setup_done = False
+        self._accumulating = False
         # RCCL supports ReduceOp.AVG; we pre-divide instead so one code path
         # covers gloo + compressors (the divide is fused into the HIP pack /
         # cast kernels on GPU).
@@ -371,14 +372,34 @@ class DistributedEngine:
 
     def _make_bucket_hook(self, bucket: Bucket):
         def hook(_param):
+            if self._accumulating:
+                return
             bucket.mark_ready_and_maybe_issue(self)
         return hook
 
     def _make_shard_hook(self, plan: VarPlan):
         def hook(_param):
+            if self._accumulating:
+                return
             for sh in plan.shards:
                 sh.reducer.issue(self)
         return hook
+
+    def no_sync(self):
+        """Context manager: skip gradient synchronization for accumulation
+        micro-batches (grads accumulate into the flat buckets); the LAST
+        backward runs outside it (DDP-style semantics)."""
+        engine = self
+
+        class _NoSync:
+            def __enter__(self):
+                engine._accumulating = True
+
+            def __exit__(self, *exc):
+                engine._accumulating = False
+                return False
+
+        return _NoSync()
 
     def _patch_optimizer(self):
         """Route the captured optimizer's step/zero_grad through the engine
