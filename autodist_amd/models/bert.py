@@ -10,7 +10,6 @@ embedding tables go to load-balanced PS owners.
 Attention uses torch.scaled_dot_product_attention (ROCm SDPA); params stay
 fp32 with bf16 autocast compute.
 """
-import math
 
 import torch
 import torch.nn as nn

@@ -17,7 +17,6 @@ from the node's actual fabric instead:
 Numbers are initialization defaults — `fit()` can overwrite them from
 measured (size -> time) samples collected by the profiler.
 """
-import math
 from typing import Dict
 
 from autodist_amd.proto.strategy_ir import CompressorType
