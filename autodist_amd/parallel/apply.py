@@ -21,7 +21,8 @@ from typing import Dict, List
 
 import torch
 
-_SUPPORTED = ("SGD", "Adam", "AdamW", "Adagrad", "RMSprop")
+_SUPPORTED = ("SGD", "Adam", "AdamW", "Adagrad", "RMSprop", "Adamax",
+              "NAdam", "RAdam", "Adadelta", "ASGD", "Rprop")
 
 
 def is_supported(cls_name: str) -> bool:
@@ -48,6 +49,33 @@ def make_state(cls_name: str, param: torch.Tensor, hyper: dict) -> Dict[str, tor
         if hyper.get("centered", False):
             s["grad_avg"] = torch.zeros_like(param)
         return s
+    if cls_name == "Adamax":
+        return {"step": torch.zeros((), dtype=torch.float32),
+                "exp_avg": torch.zeros_like(param),
+                "exp_inf": torch.zeros_like(param)}
+    if cls_name == "NAdam":
+        return {"step": torch.zeros((), dtype=torch.float32),
+                "mu_product": torch.ones((), dtype=torch.float32),
+                "exp_avg": torch.zeros_like(param),
+                "exp_avg_sq": torch.zeros_like(param)}
+    if cls_name == "RAdam":
+        return {"step": torch.zeros((), dtype=torch.float32),
+                "exp_avg": torch.zeros_like(param),
+                "exp_avg_sq": torch.zeros_like(param)}
+    if cls_name == "Adadelta":
+        return {"step": torch.zeros((), dtype=torch.float32),
+                "square_avg": torch.zeros_like(param),
+                "acc_delta": torch.zeros_like(param)}
+    if cls_name == "ASGD":
+        return {"step": torch.zeros((), dtype=torch.float32),
+                "eta": torch.tensor(float(hyper.get("lr", 1e-2))),
+                "mu": torch.ones(()),
+                "ax": param.detach().clone().to(torch.float32)}
+    if cls_name == "Rprop":
+        return {"step": torch.zeros((), dtype=torch.float32),
+                "prev": torch.zeros_like(param),
+                "step_size": torch.full_like(
+                    param, float(hyper.get("lr", 1e-2)))}
     raise NotImplementedError(f"optimizer {cls_name} not supported for engine apply")
 
 
@@ -178,8 +206,135 @@ def apply_rmsprop(params, grads, states, hyper):
             p.addcdiv_(g, a, value=-lr)
 
 
+def apply_adamax(params, grads, states, hyper):
+    lr = hyper["lr"]
+    beta1, beta2 = hyper.get("betas", (0.9, 0.999))
+    eps = hyper.get("eps", 1e-8)
+    wd = hyper.get("weight_decay", 0.0)
+    grads = _maybe_weight_decay(grads, params, wd)
+    for p, g, st in zip(params, grads, states):
+        st["step"] += 1
+        step = float(st["step"])
+        st["exp_avg"].lerp_(g, 1 - beta1)
+        torch.maximum(st["exp_inf"].mul_(beta2), g.abs().add_(eps),
+                      out=st["exp_inf"])
+        clr = lr / (1 - beta1 ** step)
+        p.addcdiv_(st["exp_avg"], st["exp_inf"], value=-clr)
+
+
+def apply_nadam(params, grads, states, hyper):
+    lr = hyper["lr"]
+    beta1, beta2 = hyper.get("betas", (0.9, 0.999))
+    eps = hyper.get("eps", 1e-8)
+    wd = hyper.get("weight_decay", 0.0)
+    psi = hyper.get("momentum_decay", 4e-3)
+    decoupled = hyper.get("decoupled_weight_decay", False)
+    if wd != 0 and decoupled:
+        torch._foreach_mul_(params, 1 - lr * wd)
+    elif wd != 0:
+        grads = torch._foreach_add(grads, params, alpha=wd)
+    for p, g, st in zip(params, grads, states):
+        st["step"] += 1
+        step = float(st["step"])
+        bc2 = 1 - beta2 ** step
+        mu = beta1 * (1.0 - 0.5 * 0.96 ** (step * psi))
+        mu_next = beta1 * (1.0 - 0.5 * 0.96 ** ((step + 1) * psi))
+        st["mu_product"] *= mu
+        mu_product = float(st["mu_product"])
+        st["exp_avg"].lerp_(g, 1 - beta1)
+        st["exp_avg_sq"].mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        denom = st["exp_avg_sq"].div(bc2).sqrt_().add_(eps)
+        p.addcdiv_(g, denom, value=-lr * (1 - mu) / (1 - mu_product))
+        p.addcdiv_(st["exp_avg"], denom,
+                   value=-lr * mu_next / (1 - mu_product * mu_next))
+
+
+def apply_radam(params, grads, states, hyper):
+    lr = hyper["lr"]
+    beta1, beta2 = hyper.get("betas", (0.9, 0.999))
+    eps = hyper.get("eps", 1e-8)
+    wd = hyper.get("weight_decay", 0.0)
+    decoupled = hyper.get("decoupled_weight_decay", False)
+    if wd != 0 and decoupled:
+        torch._foreach_mul_(params, 1 - lr * wd)
+    elif wd != 0:
+        grads = torch._foreach_add(grads, params, alpha=wd)
+    rho_inf = 2.0 / (1 - beta2) - 1
+    for p, g, st in zip(params, grads, states):
+        st["step"] += 1
+        step = float(st["step"])
+        bc1 = 1 - beta1 ** step
+        bc2 = 1 - beta2 ** step
+        st["exp_avg"].lerp_(g, 1 - beta1)
+        st["exp_avg_sq"].mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        m_hat = st["exp_avg"] / bc1
+        rho_t = rho_inf - 2 * step * (beta2 ** step) / bc2
+        if rho_t > 5.0:
+            rect = math.sqrt((rho_t - 4) * (rho_t - 2) * rho_inf
+                             / ((rho_inf - 4) * (rho_inf - 2) * rho_t))
+            adaptive_lr = math.sqrt(bc2) / st["exp_avg_sq"].sqrt().add_(eps)
+            p.add_(m_hat * lr * adaptive_lr * rect, alpha=-1.0)
+        else:
+            p.add_(m_hat * lr, alpha=-1.0)
+
+
+def apply_adadelta(params, grads, states, hyper):
+    lr = hyper["lr"]
+    rho = hyper.get("rho", 0.9)
+    eps = hyper.get("eps", 1e-6)
+    wd = hyper.get("weight_decay", 0.0)
+    grads = _maybe_weight_decay(grads, params, wd)
+    for p, g, st in zip(params, grads, states):
+        st["step"] += 1
+        st["square_avg"].mul_(rho).addcmul_(g, g, value=1 - rho)
+        std = st["square_avg"].add(eps).sqrt_()
+        delta = st["acc_delta"].add(eps).sqrt_().div_(std).mul_(g)
+        st["acc_delta"].mul_(rho).addcmul_(delta, delta, value=1 - rho)
+        p.add_(delta, alpha=-lr)
+
+
+def apply_asgd(params, grads, states, hyper):
+    lr = hyper["lr"]
+    lambd = hyper.get("lambd", 1e-4)
+    alpha = hyper.get("alpha", 0.75)
+    t0 = hyper.get("t0", 1e6)
+    wd = hyper.get("weight_decay", 0.0)
+    grads = _maybe_weight_decay(grads, params, wd)
+    for p, g, st in zip(params, grads, states):
+        st["step"] += 1
+        step = float(st["step"])
+        eta = float(st["eta"])
+        mu = float(st["mu"])
+        p.mul_(1 - lambd * eta)
+        p.add_(g, alpha=-eta)
+        if mu != 1:
+            st["ax"].add_(p.sub(st["ax"]).mul_(mu))
+        else:
+            st["ax"].copy_(p)
+        st["eta"] = torch.tensor(lr / (1 + lambd * lr * step) ** alpha)
+        st["mu"] = torch.tensor(1.0 / max(1.0, step - t0))
+
+
+def apply_rprop(params, grads, states, hyper):
+    etaminus, etaplus = hyper.get("etas", (0.5, 1.2))
+    step_min, step_max = hyper.get("step_sizes", (1e-6, 50.0))
+    for p, g, st in zip(params, grads, states):
+        st["step"] += 1
+        sign = g.mul(st["prev"]).sign()
+        mult = torch.where(sign > 0, etaplus,
+                           torch.where(sign < 0, etaminus, 1.0))
+        st["step_size"].mul_(mult).clamp_(step_min, step_max)
+        g = g.clone()
+        g[sign.eq(-1)] = 0
+        p.addcmul_(g.sign(), st["step_size"], value=-1)
+        st["prev"].copy_(g)
+
+
 _APPLY = {"SGD": apply_sgd, "Adam": apply_adam, "AdamW": apply_adamw,
-          "Adagrad": apply_adagrad, "RMSprop": apply_rmsprop}
+          "Adagrad": apply_adagrad, "RMSprop": apply_rmsprop,
+          "Adamax": apply_adamax, "NAdam": apply_nadam, "RAdam": apply_radam,
+          "Adadelta": apply_adadelta, "ASGD": apply_asgd,
+          "Rprop": apply_rprop}
 
 
 def apply_dense(cls_name: str, params, grads, states, hyper):

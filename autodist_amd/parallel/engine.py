@@ -277,8 +277,8 @@ class DistributedEngine:
             if not all_simple_ar:
                 raise NotImplementedError(
                     f"optimizer(s) {unsupported} not supported by the engine "
-                    "applier; PS/partitioned/sparse strategies need one of "
-                    f"{('SGD', 'Adam', 'AdamW', 'Adagrad', 'RMSprop')}")
+                    "applier; PS/partitioned/sparse strategies need a torch "
+                    "optimizer class with an engine apply (parallel/apply.py)")
             self._fallback_user_opt = True
             logging.info("unsupported optimizer %s: falling back to user "
                          "optimizer.step() after gradient sync", unsupported)
@@ -595,6 +595,18 @@ class DistributedEngine:
             if hyper.get("centered"):
                 keys.append("grad_avg")
             return keys
+        if cls_name == "Adamax":
+            return ["step", "exp_avg", "exp_inf"]
+        if cls_name == "NAdam":
+            return ["step", "mu_product", "exp_avg", "exp_avg_sq"]
+        if cls_name == "RAdam":
+            return ["step", "exp_avg", "exp_avg_sq"]
+        if cls_name == "Adadelta":
+            return ["step", "square_avg", "acc_delta"]
+        if cls_name == "ASGD":
+            return ["step", "eta", "mu", "ax"]
+        if cls_name == "Rprop":
+            return ["step", "prev", "step_size"]
         return []
 
     def consolidated_optimizer_state(self) -> Dict[str, dict]:

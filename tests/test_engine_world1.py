@@ -57,6 +57,14 @@ OPTS = [
     ("adamw", lambda ps: torch.optim.AdamW(ps, lr=1e-2, weight_decay=0.05)),
     ("adagrad", lambda ps: torch.optim.Adagrad(ps, lr=1e-2)),
     ("rmsprop", lambda ps: torch.optim.RMSprop(ps, lr=1e-3, momentum=0.9)),
+    ("adamax", lambda ps: torch.optim.Adamax(ps, lr=1e-2, weight_decay=1e-3)),
+    ("nadam", lambda ps: torch.optim.NAdam(ps, lr=1e-2)),
+    ("nadam_wd", lambda ps: torch.optim.NAdam(ps, lr=1e-2, weight_decay=0.02,
+                                              decoupled_weight_decay=True)),
+    ("radam", lambda ps: torch.optim.RAdam(ps, lr=1e-2)),
+    ("adadelta", lambda ps: torch.optim.Adadelta(ps, lr=0.5)),
+    ("asgd", lambda ps: torch.optim.ASGD(ps, lr=1e-2)),
+    ("rprop", lambda ps: torch.optim.Rprop(ps, lr=1e-2)),
 ]
 
 
@@ -138,13 +146,17 @@ def test_gradient_accumulation_no_sync():
         assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
 
 
+class _CustomOpt(torch.optim.SGD):
+    """A user-defined optimizer class unknown to the engine applier."""
+
+
 def test_unsupported_optimizer_fallback():
-    """NAdam isn't in the engine applier: pure-AR strategies fall back to the
-    user optimizer after gradient sync."""
+    """Unknown optimizer classes: pure-AR strategies fall back to the user
+    optimizer after gradient sync."""
     data = _data()
     m_ref = make_model()
     m_eng = copy.deepcopy(m_ref)
-    mk = lambda ps: torch.optim.NAdam(ps, lr=1e-2)  # noqa: E731
+    mk = lambda ps: _CustomOpt(ps, lr=1e-2, momentum=0.9)  # noqa: E731
     torch_train(m_ref, mk, data, len(data))
     engine_train(m_eng, mk, data, len(data), AllReduce())
     for p1, p2 in zip(m_ref.parameters(), m_eng.parameters()):
@@ -155,7 +167,7 @@ def test_unsupported_optimizer_with_ps_raises():
     g = GraphItem()
     m = make_model()
     g.extend_model(m)
-    opt = torch.optim.NAdam(m.parameters(), lr=1e-2)
+    opt = _CustomOpt(m.parameters(), lr=1e-2)
     g.extend_optimizer_info(opt)
     strategy = PS().build(g, ResourceSpec())
     engine = DistributedEngine(g, strategy, rank=0, world_size=1,
