@@ -638,6 +638,24 @@ class DistributedEngine:
                         state[k] = t.clone()
                     else:
                         state[k] = t[off:off + n].view(plan.param.shape).clone()
+            elif (plan.shards[0].kind == "local"
+                  and getattr(plan.param, "_autodist_shard_range", None)
+                  and self.world_size > 1):
+                # exclusively-owned rows (ShardedEmbedding): assemble the full
+                # per-row state across ranks (SaveSliceInfo semantics)
+                from autodist_amd.parallel.comm import allgatherv
+                sh = plan.shards[0]
+                nrows = plan.param.shape[0]
+                for k in keys:
+                    t = (sh.state or {}).get(k)
+                    if t is None:
+                        continue
+                    if t.dim() == 0 or t.shape[0] != nrows:
+                        state[k] = t.clone()
+                        continue
+                    parts = allgatherv(t.detach().contiguous(),
+                                       self.world_size, self.process_group)
+                    state[k] = torch.cat(parts, dim=0).cpu()
             elif plan.sparse or (len(plan.shards) == 1
                                  and plan.shards[0].kind in ("allreduce",
                                                              "local")
@@ -732,6 +750,22 @@ class DistributedEngine:
                         if k not in b.state:
                             b.state[k] = torch.zeros_like(b.flat)
                         b.state[k][off:off + n].view(plan.param.shape).copy_(src)
+            elif (plan.shards[0].kind == "local"
+                  and getattr(plan.param, "_autodist_shard_range", None)):
+                # slice the full per-row state back to the local rows
+                rng = plan.param._autodist_shard_range
+                sh = plan.shards[0]
+                if sh.state is None:
+                    sh.state = apply_mod.make_state(plan.cls_name,
+                                                    plan.param.data, plan.hyper)
+                for k in keys:
+                    if k not in state:
+                        continue
+                    src = state[k]
+                    if src.dim() == 0 or src.shape[0] != rng[2]:
+                        sh.state[k] = src.clone().to(self.device)
+                    else:
+                        sh.state[k] = src[rng[0]:rng[1]].clone().to(self.device)
             else:
                 axis = plan.shards[0].slice.axis if plan.shards[0].slice else 0
                 for sh in plan.shards:

@@ -133,13 +133,51 @@ class ShardedEmbedding(torch.nn.Module):
         bounds = split_boundaries(num_embeddings, self.world_size)
         self.row_start, self.row_end = bounds[self.rank]
         # boundaries for bucketize: end of each shard except the last
+        # (non-persistent: the state_dict carries the consolidated table only)
         self.register_buffer("boundaries", torch.tensor(
-            [e for (_, e) in bounds[:-1]], dtype=torch.int64))
+            [e for (_, e) in bounds[:-1]], dtype=torch.int64),
+            persistent=False)
         shard = torch.empty(self.row_end - self.row_start, embedding_dim)
         torch.nn.init.normal_(shard, std=0.01)
         self.shard = torch.nn.Parameter(shard)
         # each rank exclusively owns its rows: no gradient sync needed
         self.shard._autodist_shard_local = True
+        # checkpoint consolidation metadata (engine/saver use this to
+        # reassemble full optimizer state — reference SaveSliceInfo analog)
+        self.shard._autodist_shard_range = (self.row_start, self.row_end,
+                                            num_embeddings)
+        self.shard._autodist_shard_module = self
+
+    # -- checkpoint integration: state_dict is nn.Embedding-compatible ------
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        """Emit the CONSOLIDATED table as '<prefix>weight' so checkpoints are
+        interchangeable with nn.Embedding (collective: every rank must be
+        saving, which the AutoDist Saver guarantees)."""
+        destination[prefix + "weight"] = self.full_weight().cpu()
+
+    def _load_from_state_dict(self, state_dict, prefix, local_metadata,
+                              strict, missing_keys, unexpected_keys,
+                              error_msgs):
+        key = prefix + "weight"
+        if key in state_dict:
+            full = state_dict[key]
+            if tuple(full.shape) == (self.num_embeddings, self.dim):
+                with torch.no_grad():
+                    self.shard.copy_(full[self.row_start:self.row_end])
+            else:
+                error_msgs.append(
+                    f"ShardedEmbedding {key}: expected "
+                    f"{(self.num_embeddings, self.dim)}, got {tuple(full.shape)}")
+            state_dict = dict(state_dict)
+            del state_dict[key]
+        elif strict:
+            missing_keys.append(key)
+        # also accept a raw per-rank 'shard' entry (legacy/per-rank ckpt)
+        skey = prefix + "shard"
+        if skey in state_dict and tuple(state_dict[skey].shape) == \
+                tuple(self.shard.shape):
+            with torch.no_grad():
+                self.shard.copy_(state_dict[skey])
 
     def forward(self, ids):
         return _ShardedLookup.apply(self.shard, ids, self)

@@ -153,12 +153,15 @@ def test_sharded_embedding_gloo():
     run_distributed(_sharded_embedding_case, world_size=2)
 
 
-def _sharded_ncf_case(rank, world):
+def _sharded_ncf_case(rank, world, tmpdir):
+    from autodist_amd.checkpoint.saver import Saver
     from autodist_amd.models.ncf import NeuMF
     torch.manual_seed(3)
     model = NeuMF(40, 60, mf_dim=8, mlp_dims=(16, 8), sharded=True)
-    opt, engine = _engine_for(model, lambda p: torch.optim.SGD(p, lr=0.05),
-                              AllReduce(), world=world, rank=rank)
+    g = None
+    opt, engine = _engine_for(model, lambda p: torch.optim.SGD(
+        p, lr=0.05, momentum=0.9), AllReduce(), world=world, rank=rank)
+    saver = Saver(graph_item=engine.graph_item)
     for s in range(3):
         torch.manual_seed(200 + 10 * s + rank)
         u = torch.randint(0, 40, (16,))
@@ -175,9 +178,28 @@ def _sharded_ncf_case(rank, world):
         lst = [torch.zeros_like(p) for _ in range(world)]
         dist.all_gather(lst, p.detach())
         assert torch.allclose(lst[0], lst[1], atol=1e-6), name
+    # checkpoint: consolidated, nn.Embedding-compatible
+    path = saver.save(tmpdir + "/sharded_ckpt")
+    dist.barrier()
+    ckpt = torch.load(path, weights_only=False)
+    assert tuple(ckpt["model"]["mf_user.weight"].shape) == (40, 8)
+    vanilla = NeuMF(40, 60, mf_dim=8, mlp_dims=(16, 8), sparse=True)
+    vanilla.load_state_dict(ckpt["model"])
+    assert torch.allclose(vanilla.mf_user.weight.detach(),
+                          model.mf_user.full_weight().cpu(), atol=1e-6)
+    # restore into a fresh sharded model: shard + momentum state round-trip
+    torch.manual_seed(99)
+    model2 = NeuMF(40, 60, mf_dim=8, mlp_dims=(16, 8), sharded=True)
+    opt2, engine2 = _engine_for(model2, lambda p: torch.optim.SGD(
+        p, lr=0.05, momentum=0.9), AllReduce(), world=world, rank=rank)
+    saver2 = Saver(graph_item=engine2.graph_item)
+    saver2.restore(path)
+    assert torch.allclose(model2.mf_user.shard.detach(),
+                          model.mf_user.shard.detach(), atol=1e-6)
     engine.teardown()
+    engine2.teardown()
 
 
 @pytest.mark.integration
-def test_sharded_ncf_gloo():
-    run_distributed(_sharded_ncf_case, world_size=2)
+def test_sharded_ncf_gloo(tmp_path):
+    run_distributed(_sharded_ncf_case, world_size=2, args=(str(tmp_path),))
