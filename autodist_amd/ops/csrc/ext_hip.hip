@@ -8,6 +8,7 @@
 #include "multi_tensor.hip"
 #include "bn_ops.hip"
 #include "psgd_gemm.hip"
+#include "mfma_probe.hip"
 
 namespace {
 
@@ -415,6 +416,17 @@ void psgd_decompress_ef(torch::Tensor flat, torch::Tensor err,
                      Qp.data_ptr<float>(), numel, s, (float)scale);
 }
 
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B, long cand) {
+  TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32);
+  TORCH_CHECK(B.scalar_type() == torch::kBFloat16 && B.numel() == 32 * 16);
+  auto D = torch::zeros({16, 16}, A.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(A.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(B.data_ptr()),
+                     D.data_ptr<float>(), (int)cand);
+  return D;
+}
+
 void psgd_add_err_pad(torch::Tensor flat, torch::Tensor err,
                       torch::Tensor out) {
   long numel = flat.numel();
@@ -448,4 +460,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused hat = P Q^T * scale; err = M - hat; flat = hat");
   m.def("psgd_add_err_pad", &psgd_add_err_pad,
         "padded M = flat + err (zero tail)");
+  m.def("mfma_probe", &mfma_probe,
+        "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
 }

@@ -1,0 +1,34 @@
+"""Diagnostic (always-passing) probe of the gfx950 bf16 MFMA fragment
+layout — the round-2 attention kernel starts from whatever candidate this
+confirms. Results land in the test log."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_mfma_bf16_layout_probe(capsys):
+    from autodist_amd.ops import api
+    assert api.has_gpu_ops()
+    ext = api.ext()
+    torch.manual_seed(0)
+    # ASYMMETRIC operands (guide: symmetric B passes transposed layouts)
+    A = (torch.arange(16 * 32, device="cuda").float() % 7 - 3).view(16, 32)
+    B = (torch.arange(32 * 16, device="cuda").float() % 5 - 2).view(32, 16)
+    B[3, 11] = 9.0  # extra asymmetry
+    ref = (A @ B)
+    verdicts = {}
+    for cand in range(3):
+        D = ext.mfma_probe(A.to(torch.bfloat16).contiguous(),
+                           B.to(torch.bfloat16).contiguous(), cand)
+        err = (D - ref).abs().max().item()
+        verdicts[cand] = err
+        print(f"MFMA-PROBE candidate {cand}: max err {err:.4f} "
+              f"{'MATCH' if err < 1.0 else 'mismatch'}")
+    best = min(verdicts, key=verdicts.get)
+    print(f"MFMA-PROBE best candidate: {best} (err {verdicts[best]:.4f})")
+    with capsys.disabled():
+        print(f"\n[MFMA-PROBE] v_mfma_f32_16x16x32_bf16 A/B layout: "
+              f"candidate {best} err={verdicts[best]:.4f} "
+              f"(0=contig8, 1=stride4, 2=two-4-blocks)")
+    # diagnostic only: never fail the suite on a layout surprise
