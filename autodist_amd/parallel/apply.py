@@ -364,7 +364,11 @@ def apply_sparse_rows(cls_name: str, param: torch.Tensor, rows: torch.Tensor,
     Matches torch.optim sparse semantics (SGD/Adagrad support sparse grads;
     sparse Adam follows torch.optim.SparseAdam)."""
     if cls_name == "SGD":
-        # torch SGD with sparse grad ignores momentum/wd must be 0
+        # torch SGD rejects sparse grads with momentum/weight_decay; match it
+        if hyper.get("momentum", 0) or hyper.get("weight_decay", 0):
+            raise ValueError(
+                "sparse SGD does not support momentum/weight_decay "
+                "(torch.optim.SGD semantics)")
         param.index_add_(0, rows, row_grads, alpha=-hyper["lr"])
         return
     if cls_name == "Adagrad":

@@ -73,8 +73,9 @@ def _case(rank, world, kind, strat_name, strat_kwargs):
         assert torch.isfinite(p).all(), name
         lst = [torch.zeros_like(p) for _ in range(world)]
         dist.all_gather(lst, p.detach())
-        assert torch.allclose(lst[0], lst[1], atol=1e-5), \
-            f"{name}: replicas diverged by {(lst[0]-lst[1]).abs().max()}"
+        for other in lst[1:]:
+            assert torch.allclose(lst[0], other, atol=1e-5), \
+                f"{name}: replicas diverged by {(lst[0]-other).abs().max()}"
     engine.teardown()
 
 
@@ -82,3 +83,11 @@ def _case(rank, world, kind, strat_name, strat_kwargs):
 @pytest.mark.parametrize("strat_name,strat_kwargs", STRATEGIES)
 def test_matrix(kind, strat_name, strat_kwargs):
     run_distributed(_case, world_size=2, args=(kind, strat_name, strat_kwargs))
+
+
+@pytest.mark.parametrize("strat_name", ["PartitionedPS", "PSLoadBalancing",
+                                        "AllReduce", "Parallax"])
+def test_matrix_world4(strat_name):
+    """Spot-check at world_size=4 (multiple PS owners per var; deeper rank
+    fan-out than the standard 2-rank cases)."""
+    run_distributed(_case, world_size=4, args=("mlp", strat_name, {}))
