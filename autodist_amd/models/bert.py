@@ -54,9 +54,11 @@ class BertSelfAttention(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        o = torch.nn.functional.scaled_dot_product_attention(
-            q, k, v, attn_mask=attn_mask,
-            dropout_p=self.dropout if self.training else 0.0)
+        # serving path: the fused gfx950 MFMA attention kernel (falls back to
+        # torch SDPA for training / masked / non-bf16 cases)
+        from autodist_amd.ops.fused_attention import fused_sdpa
+        o = fused_sdpa(q, k, v, attn_mask=attn_mask,
+                       dropout_p=self.dropout if self.training else 0.0)
         o = o.transpose(1, 2).reshape(B, S, H)
         return self.out(o)
 

@@ -9,6 +9,7 @@
 #include "bn_ops.hip"
 #include "psgd_gemm.hip"
 #include "mfma_probe.hip"
+#include "attention.hip"
 
 namespace {
 
@@ -416,6 +417,26 @@ void psgd_decompress_ef(torch::Tensor flat, torch::Tensor err,
                      Qp.data_ptr<float>(), numel, s, (float)scale);
 }
 
+torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                       double scale) {
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == ATTN_D,
+              "attn_fwd expects [B,H,S,64]");
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  long B = q.size(0), H = q.size(1), S = q.size(2);
+  TORCH_CHECK(S % 32 == 0 && S >= 32, "S must be a multiple of 32");
+  TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes());
+  auto o = torch::empty_like(q);
+  dim3 grid(S / 16, B * H);
+  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(64), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()), S,
+                     (float)scale);
+  return o;
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B, long cand) {
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32);
   TORCH_CHECK(B.scalar_type() == torch::kBFloat16 && B.numel() == 32 * 16);
@@ -462,4 +483,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "padded M = flat + err (zero tail)");
   m.def("mfma_probe", &mfma_probe,
         "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
+  m.def("attn_fwd", &attn_fwd,
+        "fused MFMA attention forward (bf16, D=64, no mask) — serving path");
 }
