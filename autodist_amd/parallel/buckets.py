@@ -12,7 +12,6 @@ sized for the xGMI mesh (7 p2p links x ~153 GB/s per GPU).
 from typing import Dict, List, Optional
 
 import torch
-import torch.distributed as dist
 
 from autodist_amd.const import DEFAULT_BUCKET_BYTES
 from autodist_amd.parallel.compressor import Compressor
@@ -116,15 +115,11 @@ class Bucket:
         if engine.device.type == "cuda":
             with torch.cuda.stream(engine.comm_stream):
                 self.compressor.finalize(self.flat, self._handle)
-                if engine.avg_supported_needs_post_div:
-                    self.flat.mul_(1.0 / engine.world_size)
             self.done_event = torch.cuda.Event()
             self.done_event.record(engine.comm_stream)
             torch.cuda.current_stream().wait_event(self.done_event)
         else:
             self.compressor.finalize(self.flat, self._handle)
-            if engine.avg_supported_needs_post_div:
-                self.flat.mul_(1.0 / engine.world_size)
 
 
 def build_buckets(items, device: torch.device,
@@ -147,8 +142,11 @@ def build_buckets(items, device: torch.device,
         key = (group_id, param.dtype, cls_name, hyper_key)
         by_group.setdefault(key, []).append((param, comp_type, hyper))
     buckets: List[Bucket] = []
+    # sort key stringifies the hyper tuple: values of mixed types (None vs
+    # bool from torch param-group internals) are not mutually orderable
     for (group_id, dtype, cls_name, _), members in sorted(
-            by_group.items(), key=lambda kv: (kv[0][0], kv[0][2], kv[0][3])):
+            by_group.items(),
+            key=lambda kv: (kv[0][0], kv[0][2], repr(kv[0][3]))):
         members = list(reversed(members))
         current = None
         elt = torch.empty((), dtype=dtype).element_size()

@@ -178,11 +178,9 @@ class DistributedEngine:
         self._fallback_user_opt = False
         self._setup_done = False
         self._accumulating = False
-        # RCCL supports ReduceOp.AVG; we pre-divide instead so one code path
-        # covers gloo + compressors (the divide is fused into the HIP pack /
-        # cast kernels on GPU).
-        self.avg_supported = False
-        self.avg_supported_needs_post_div = False
+        # note: RCCL supports ReduceOp.AVG, but the mean is instead fused as
+        # a scale into the compress/cast kernels (one code path for gloo +
+        # every compressor)
 
     # ------------------------------------------------------------------ set-up
     def maybe_init_process_group(self):

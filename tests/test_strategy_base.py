@@ -149,6 +149,26 @@ def test_auto_strategy_picks_something(tmp_gpu_resource_spec):
     assert len(s.node_config) == 2
 
 
+def test_compiler_device_resolver():
+    """Device-string resolution hook (reference DeviceResolver,
+    kernel/device/resolver.py:47-67: "ip:GPU:k" -> runtime device)."""
+    from autodist_amd.strategy.base import StrategyCompiler
+    g = _graph_item(shapes=[(8, 4)])
+    rs = ResourceSpec()
+    s = PS().build(g, rs)
+    s.graph_config.replicas = ["10.0.0.1:GPU:0", "10.0.0.1:GPU:1"]
+    s.node_config[0].ps_synchronizer.reduction_destination = "10.0.0.1:GPU:1"
+
+    def resolver(dev):  # ip:GPU:k -> rank string
+        return f"rank:{dev.rsplit(':', 1)[1]}"
+
+    out = StrategyCompiler(g).set_device_resolver(resolver).compile(s)
+    assert out.graph_config.replicas == ["rank:0", "rank:1"]
+    assert out.node_config[0].ps_synchronizer.reduction_destination == "rank:1"
+    # original strategy untouched (compile works on a copy)
+    assert s.graph_config.replicas[0] == "10.0.0.1:GPU:0"
+
+
 def test_compiler_prunes_unknown_vars():
     from autodist_amd.strategy.base import StrategyCompiler
     g = _graph_item(shapes=[(4, 4)])
