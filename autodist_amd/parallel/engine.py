@@ -799,6 +799,37 @@ class DistributedEngine:
     def step_count(self) -> int:
         return self._step_count
 
+    def stats(self) -> dict:
+        """Per-step synchronization accounting (observability — the
+        reference had logging only, §5.5)."""
+        ar_bytes = sum(b.nbytes for b in self.buckets)
+        ps_bytes = 0
+        shard_reduce_bytes = 0
+        n_ps_shards = 0
+        n_reducers = 0
+        for plan in self.var_plans:
+            for sh in plan.shards:
+                size = (sh.slice.view(plan.param.data).numel() if sh.slice
+                        else plan.param.numel()) * plan.param.element_size()
+                if sh.kind == "ps":
+                    ps_bytes += size
+                    n_ps_shards += 1
+                elif sh.reducer is not None:
+                    shard_reduce_bytes += size
+                    n_reducers += 1
+        return {
+            "step_count": self._step_count,
+            "world_size": self.world_size,
+            "n_buckets": len(self.buckets),
+            "allreduce_bytes_per_step": ar_bytes,
+            "ps_shards": n_ps_shards,
+            "ps_bytes_per_step": ps_bytes,
+            "partitioned_ar_shards": n_reducers,
+            "partitioned_ar_bytes_per_step": shard_reduce_bytes,
+            "sparse_vars": sum(1 for p in self.var_plans if p.sparse),
+            "fallback_user_optimizer": self._fallback_user_opt,
+        }
+
     def teardown(self):
         for h in self._hook_handles:
             h.remove()

@@ -55,6 +55,31 @@ def test_cost_model_fit():
         cm.ps_round_trip_time(100e6, 8, owners=1)
 
 
+def test_engine_stats():
+    import torch
+    from autodist_amd.graph_item import GraphItem
+    from autodist_amd.parallel.engine import DistributedEngine
+    from autodist_amd.resource_spec import ResourceSpec
+    from autodist_amd.strategy import Parallax
+    torch.manual_seed(0)
+    emb = torch.nn.Embedding(10, 4, sparse=True)
+    lin = torch.nn.Linear(4, 2)
+    model = torch.nn.ModuleDict({"e": emb, "l": lin})
+    g = GraphItem()
+    g.extend_model(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    g.extend_optimizer_info(opt)
+    engine = DistributedEngine(g, Parallax().build(g, ResourceSpec()),
+                               rank=0, world_size=1,
+                               device=torch.device("cpu")).setup()
+    s = engine.stats()
+    assert s["n_buckets"] >= 1
+    assert s["allreduce_bytes_per_step"] == (4 * 2 + 2) * 4  # lin w+b fp32
+    assert s["sparse_vars"] == 1
+    assert not s["fallback_user_optimizer"]
+    engine.teardown()
+
+
 @pytest.mark.integration
 def test_coordinator_fail_fast(tmp_path):
     """A crashing worker must bring down the chief (reference
