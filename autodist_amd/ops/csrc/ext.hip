@@ -9,6 +9,7 @@
 #include "psgd_gemm.hip"
 #include "mfma_probe.hip"
 #include "attention.hip"
+#include "attention_bwd.hip"
 
 namespace {
 
@@ -436,6 +437,35 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return o;
 }
 
+std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, torch::Tensor o,
+                                    torch::Tensor dout, double scale) {
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == ATTN_BD);
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  long B = q.size(0), H = q.size(1), S = q.size(2);
+  TORCH_CHECK(S % 32 == 0 && S >= 32);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(q);
+  auto dv = torch::empty_like(q);
+  auto fopt = q.options().dtype(torch::kFloat32);
+  auto Mbuf = torch::empty({B * H * S}, fopt);
+  auto Lbuf = torch::empty({B * H * S}, fopt);
+  auto Dbuf = torch::empty({B * H * S}, fopt);
+  dim3 grid(S / 16, B * H);
+  auto st = cur_stream();
+#define BF16P(t) reinterpret_cast<__hip_bfloat16*>((t).data_ptr())
+  hipLaunchKernelGGL(attn_bwd_q_kernel, grid, dim3(64), 0, st, BF16P(q),
+                     BF16P(k), BF16P(v), BF16P(o), BF16P(dout), BF16P(dq),
+                     Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
+                     Dbuf.data_ptr<float>(), S, (float)scale);
+  hipLaunchKernelGGL(attn_bwd_kv_kernel, grid, dim3(64), 0, st, BF16P(q),
+                     BF16P(k), BF16P(v), BF16P(dout), BF16P(dk), BF16P(dv),
+                     Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
+                     Dbuf.data_ptr<float>(), S, (float)scale);
+#undef BF16P
+  return {dq, dk, dv};
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B, long cand) {
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32);
   TORCH_CHECK(B.scalar_type() == torch::kBFloat16 && B.numel() == 32 * 16);
@@ -484,4 +514,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
   m.def("attn_fwd", &attn_fwd,
         "fused MFMA attention forward (bf16, D=64, no mask) — serving path");
+  m.def("attn_bwd", &attn_bwd,
+        "fused MFMA attention backward (DRAFT; validate before use) -> "
+        "dq, dk, dv");
 }
