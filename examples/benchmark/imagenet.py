@@ -16,13 +16,22 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.dirname(
 
 from autodist_amd import AutoDist
 from autodist_amd import strategy as strat
-from autodist_amd.models import resnet
+from autodist_amd.models import densenet, inception, resnet, vgg
+
+MODELS = {
+    "resnet18": resnet.resnet18,
+    "resnet50": resnet.resnet50,
+    "resnet101": resnet.resnet101,
+    "vgg16": lambda **kw: vgg.vgg16(batch_norm=True, **kw),
+    "densenet121": densenet.densenet121,
+    "inception_v3": inception.inception_v3,
+}
 
 
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--model", default="resnet50",
-                        choices=["resnet18", "resnet50", "resnet101"])
+                        choices=sorted(MODELS))
     parser.add_argument("--autodist_strategy", default="AllReduce")
     parser.add_argument("--batch-size", type=int, default=64)
     parser.add_argument("--steps", type=int, default=10)
@@ -33,7 +42,7 @@ def main():
     ad = AutoDist(strategy_builder=getattr(strat, args.autodist_strategy)())
     with ad.scope():
         torch.manual_seed(0)
-        model = getattr(resnet, args.model)(fused=use_cuda)
+        model = MODELS[args.model](fused=use_cuda)
         optimizer = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                                     weight_decay=1e-4)
 

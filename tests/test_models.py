@@ -110,6 +110,33 @@ def test_ncf_partitioned_ps():
     engine.teardown()
 
 
+@pytest.mark.parametrize("maker,size", [
+    ("vgg16", 64), ("densenet121", 64), ("inception_v3", 299)])
+def test_imagenet_cnn_families_train(maker, size):
+    """VGG16 / DenseNet121 / InceptionV3 (reference benchmark models,
+    examples/benchmark/imagenet.py) run a training step through the engine."""
+    from autodist_amd.models.densenet import densenet121
+    from autodist_amd.models.inception import inception_v3
+    from autodist_amd.models.vgg import vgg16
+    makers = {"vgg16": lambda: vgg16(num_classes=10, batch_norm=True),
+              "densenet121": lambda: densenet121(num_classes=10),
+              "inception_v3": lambda: inception_v3(num_classes=10)}
+    torch.manual_seed(0)
+    model = makers[maker]()
+    opt, engine = _engine_for(model, lambda p: torch.optim.SGD(p, lr=0.01),
+                              AllReduce())
+    x = torch.randn(2, 3, size, size)
+    y = torch.randint(0, 10, (2,))
+    opt.zero_grad()
+    loss = torch.nn.functional.cross_entropy(model(x), y)
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss)
+    for p in model.parameters():
+        assert torch.isfinite(p).all()
+    engine.teardown()
+
+
 # ----------------------------------------------------- sharded embedding
 
 def _sharded_embedding_case(rank, world):
