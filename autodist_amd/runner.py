@@ -25,6 +25,23 @@ class WrappedSession:
         self.remapper = remapper
         self.graph_item = graph_item
         self._run_count = 0
+        # graph-mutation detection under AUTODIST_IS_TESTING (reference
+        # autodist.py:152-165): adding params after build means they are
+        # missing from the synchronization plan
+        from autodist_amd.const import ENV
+        self._check_mutation = ENV.AUTODIST_IS_TESTING.val
+        self._built_param_count = sum(
+            sum(1 for _ in m.parameters()) for m in graph_item.models)
+
+    def _assert_not_mutated(self):
+        count = sum(sum(1 for _ in m.parameters())
+                    for m in self.graph_item.models)
+        if count != self._built_param_count:
+            raise RuntimeError(
+                f"model mutated after the distributed session was built "
+                f"({self._built_param_count} -> {count} params): new "
+                f"parameters have no synchronization plan (reference "
+                f"autodist.py:152-165)")
 
     def run(self, fetches, feed_dict=None, options=None):
         """Execute one step.
@@ -35,6 +52,8 @@ class WrappedSession:
         feed_dict: {name: global-batch array/tensor} split across ranks.
         options: {"trace": True} dumps a chrome trace for this step.
         """
+        if self._check_mutation:
+            self._assert_not_mutated()
         feeds = self.remapper.remap_feed_dict(feed_dict)
         trace = bool(options and options.get("trace"))
         if trace:

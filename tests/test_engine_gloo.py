@@ -340,3 +340,53 @@ def _feed_fetch_case(rank, world):
 
 def test_remapper_feed_fetch():
     run_distributed(_feed_fetch_case, world_size=2)
+
+
+def _control_flow_case(rank, world):
+    """c4-analog: data-dependent control flow in forward (loop count varies
+    per step, same on every rank) — hooks/buckets must stay consistent."""
+    import torch.distributed as dist
+    from autodist_amd.graph_item import GraphItem
+    from autodist_amd.parallel.engine import DistributedEngine
+    from autodist_amd.resource_spec import ResourceSpec
+    from autodist_amd.strategy import AllReduce
+
+    class LoopNet(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.cell = torch.nn.Linear(4, 4)
+            self.head = torch.nn.Linear(4, 1)
+
+        def forward(self, x, n_steps):
+            for _ in range(n_steps):          # data-dependent depth
+                x = torch.tanh(self.cell(x))
+            return self.head(x)
+
+    torch.manual_seed(0)
+    model = LoopNet()
+    g = GraphItem()
+    g.extend_model(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    g.extend_optimizer_info(opt)
+    strategy = AllReduce().build(g, ResourceSpec())
+    strategy.graph_config.replicas = [f"127.0.0.1:CPU:{r}" for r in range(world)]
+    engine = DistributedEngine(g, strategy, rank=rank, world_size=world,
+                               device=torch.device("cpu")).setup()
+    for step in range(4):
+        torch.manual_seed(10 + 5 * step + rank)
+        x = torch.randn(6, 4)
+        y = torch.randn(6, 1)
+        opt.zero_grad()
+        # loop depth varies by STEP (identical across ranks)
+        loss = torch.nn.functional.mse_loss(model(x, 1 + step % 3), y)
+        loss.backward()
+        opt.step()
+    for p in model.parameters():
+        lst = [torch.zeros_like(p) for _ in range(world)]
+        dist.all_gather(lst, p.detach())
+        assert torch.allclose(lst[0], lst[1], atol=1e-6)
+    engine.teardown()
+
+
+def test_control_flow_model():
+    run_distributed(_control_flow_case, world_size=2)

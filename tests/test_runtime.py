@@ -81,6 +81,33 @@ def test_autodist_end_to_end_world1(monkeypatch):
     _reset_default_autodist_for_tests()
 
 
+def test_graph_mutation_detected(monkeypatch):
+    """Adding params after build is caught under AUTODIST_IS_TESTING
+    (reference autodist.py:152-165)."""
+    monkeypatch.setenv("AUTODIST_IS_TESTING", "True")
+    from autodist_amd import AutoDist
+    from autodist_amd.autodist import _reset_default_autodist_for_tests
+    from autodist_amd.strategy import AllReduce
+    _reset_default_autodist_for_tests()
+    ad = AutoDist(strategy_builder=AllReduce(), world_size=1)
+    with ad.scope():
+        model = torch.nn.Sequential(torch.nn.Linear(2, 2))
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    sess = ad.create_distributed_session()
+
+    def step(x):
+        opt.zero_grad()
+        model(x).square().mean().backward()
+        opt.step()
+        return torch.tensor(0.0)
+
+    sess.run(step, feed_dict={"x": np.ones((2, 2), np.float32)})
+    model.append(torch.nn.Linear(2, 2))  # mutate after build
+    with pytest.raises(RuntimeError, match="mutated"):
+        sess.run(step, feed_dict={"x": np.ones((2, 2), np.float32)})
+    _reset_default_autodist_for_tests()
+
+
 def test_one_autodist_per_process(monkeypatch):
     """Reference invariant (autodist.py:46-51 / tests/test_autodist.py)."""
     monkeypatch.delenv("AUTODIST_IS_TESTING", raising=False)
