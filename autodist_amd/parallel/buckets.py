@@ -13,7 +13,7 @@ from typing import Dict, List, Optional
 
 import torch
 
-from autodist_amd.const import DEFAULT_BUCKET_BYTES
+from autodist_amd.const import DEFAULT_BUCKET_BYTES, DEFAULT_FIRST_BUCKET_BYTES
 from autodist_amd.parallel.compressor import Compressor
 from autodist_amd.utils import logging
 
@@ -45,6 +45,10 @@ class Bucket:
         self._handle = None
         self.done_event: Optional[torch.cuda.Event] = None
         self._issued = False
+        # deterministic cross-rank issue key (engine assigns the md5-derived
+        # value from member var names, parallel/collective_key.py)
+        self.instance_key = bucket_id
+        self.group_key = 0
 
     def add(self, param: torch.nn.Parameter):
         self.params.append(param)
@@ -69,6 +73,25 @@ class Bucket:
     def nbytes(self) -> int:
         return self.numel * self.flat.element_size() if self.flat is not None else 0
 
+    def ensure_views(self):
+        """Re-install flat-buffer grad views if user code detached them.
+
+        model.zero_grad(set_to_none=True) bypasses the engine's patched
+        zero_grad and sets .grad = None; the next backward then allocates a
+        FRESH grad tensor that does not alias the flat buffer. Without this
+        check the engine would all-reduce and apply the stale flat contents
+        (silent corruption). Values are preserved by copying the fresh grad
+        into the view before re-pointing."""
+        for p, off, shape in zip(self.params, self.offsets, self.shapes):
+            n = p.numel()
+            view = self.flat[off:off + n].view(shape)
+            g = p.grad
+            if g is None:
+                p.grad = view
+            elif g.data_ptr() != view.data_ptr():
+                view.copy_(g.detach())
+                p.grad = view
+
     # -- per-step protocol -------------------------------------------------
     def reset(self):
         self._ready = 0
@@ -90,6 +113,7 @@ class Bucket:
 
     def issue(self, engine):
         self._issued = True
+        self.ensure_views()
         if engine.world_size <= 1:
             return
         if engine.device.type == "cuda":
@@ -106,7 +130,7 @@ class Bucket:
     def _reduce(self, engine):
         self._handle = self.compressor.reduce(
             self.flat, group=engine.process_group, async_op=True,
-            scale=1.0 / engine.world_size)
+            scale=engine.grad_scale())
 
     def finalize(self, engine):
         """Make the compute stream depend on this bucket's reduced result."""
@@ -123,35 +147,42 @@ class Bucket:
 
 
 def build_buckets(items, device: torch.device,
-                  bucket_bytes: int = DEFAULT_BUCKET_BYTES) -> List[Bucket]:
+                  bucket_bytes: int = DEFAULT_BUCKET_BYTES,
+                  first_bucket_bytes: int = DEFAULT_FIRST_BUCKET_BYTES
+                  ) -> List[Bucket]:
     """Group (param, group_id, compressor_type, cls_name, hyper, hyper_key)
     tuples into Buckets.
 
     Keeps the strategy's group ids (one ScopedAllocator-group == >=1 buckets),
     splitting any group larger than bucket_bytes so collectives overlap with
     backward instead of waiting for one giant buffer. Buckets are homogeneous
-    in (dtype, optimizer class, hyperparams) so the whole bucket updates with
-    one fused kernel.
+    in (dtype, optimizer class, hyperparams, compressor) so the whole bucket
+    updates with one fused kernel and one compressor instance.
 
     Bucket order follows REVERSED registration order within each group, since
     autograd produces gradients roughly in reverse forward order — the first
-    bucket to fill is the one holding the last layers.
+    bucket to fill is the one holding the last layers. That first bucket is
+    capped at first_bucket_bytes (DDP-style) so its collective launches early
+    in backward instead of waiting for a full-size buffer.
     """
     by_group: Dict[tuple, list] = {}
     for param, group_id, comp_type, cls_name, hyper, hyper_key in items:
-        key = (group_id, param.dtype, cls_name, hyper_key)
+        key = (group_id, param.dtype, cls_name, hyper_key, comp_type)
         by_group.setdefault(key, []).append((param, comp_type, hyper))
     buckets: List[Bucket] = []
     # sort key stringifies the hyper tuple: values of mixed types (None vs
     # bool from torch param-group internals) are not mutually orderable
-    for (group_id, dtype, cls_name, _), members in sorted(
+    for (group_id, dtype, cls_name, _, comp_type), members in sorted(
             by_group.items(),
-            key=lambda kv: (kv[0][0], kv[0][2], repr(kv[0][3]))):
+            key=lambda kv: (kv[0][0], kv[0][2], repr(kv[0][3]),
+                            repr(kv[0][4]))):
         members = list(reversed(members))
         current = None
+        cap = 0
         elt = torch.empty((), dtype=dtype).element_size()
         for param, comp_type, hyper in members:
-            if current is None or current.numel * elt >= bucket_bytes:
+            if current is None or current.numel * elt >= cap:
+                cap = first_bucket_bytes if not buckets else bucket_bytes
                 comp = Compressor.create(comp_type, f"bucket{len(buckets)}")
                 current = Bucket(len(buckets), dtype, device, comp,
                                  cls_name=cls_name, hyper=hyper)

@@ -64,7 +64,11 @@ class VarPlan:
     param: torch.nn.Parameter
     sparse: bool
     cls_name: str                  # optimizer class
-    hyper: dict                    # optimizer hyperparams for this param
+    hyper: dict                    # LIVE optimizer hyperparams (shared per
+                                   # param_group; refreshed every step so LR
+                                   # schedulers take effect — see
+                                   # engine._refresh_hyper)
+    group_index: int = 0           # index into optimizer.param_groups
     shards: List[ShardPlan] = dataclasses.field(default_factory=list)
     bucketed: bool = False
 
@@ -124,7 +128,7 @@ class ShardReducer:
         t = self._shard_grad()
         self._handle = self.compressor.reduce(t, group=engine.process_group,
                                               async_op=True,
-                                              scale=1.0 / engine.world_size)
+                                              scale=engine.grad_scale())
         self._reduced_tensor = t
 
     def finalize(self, engine):
@@ -178,6 +182,13 @@ class DistributedEngine:
         self._fallback_user_opt = False
         self._setup_done = False
         self._accumulating = False
+        # one SHARED live hyperparam dict per optimizer param_group; plans,
+        # shards and buckets all alias these dicts so a per-step refresh
+        # propagates LR-scheduler / manual param_groups changes everywhere
+        self._group_hyper: Dict[int, dict] = {}
+        # per-rank gradient weight for uneven batch splits: weighted average
+        # sum_r (n_r / N) * g_r (reference c0.py:92-119). None => 1/world.
+        self._batch_fraction: Optional[float] = None
         # note: RCCL supports ReduceOp.AVG, but the mean is instead fused as
         # a scale into the compress/cast kernels (one code path for gloo +
         # every compressor)
@@ -230,17 +241,56 @@ class DistributedEngine:
         return 0
 
     def _hyper_for(self, param) -> tuple:
+        """(cls_name, live hyper dict, param_group index) for one param.
+
+        The returned dict is SHARED by every param of the same group and is
+        refreshed from the live optimizer each step (_refresh_hyper), so LR
+        schedulers and manual param_groups edits keep torch.optim semantics."""
         opt_item = self.graph_item.optimizers[0] if self.graph_item.optimizers else None
         if opt_item is None:
             raise RuntimeError("no optimizer captured — build one inside scope()")
         opt = opt_item.optimizer
         if opt is not None:
-            for group in opt.param_groups:
+            for gidx, group in enumerate(opt.param_groups):
                 for p in group["params"]:
                     if p is param:
-                        return opt_item.cls_name, {
-                            k: v for k, v in group.items() if k != "params"}
-        return opt_item.cls_name, dict(opt_item.defaults)
+                        if gidx not in self._group_hyper:
+                            self._group_hyper[gidx] = {
+                                k: v for k, v in group.items()
+                                if k != "params"}
+                        return opt_item.cls_name, self._group_hyper[gidx], gidx
+        if -1 not in self._group_hyper:
+            self._group_hyper[-1] = dict(opt_item.defaults)
+        return opt_item.cls_name, self._group_hyper[-1], -1
+
+    def _refresh_hyper(self):
+        """Re-read param_group hyperparams from the live optimizer (in place,
+        so plan/shard/bucket aliases see the update). Fixes silent divergence
+        under LR schedulers now that opt.step() routes through the engine."""
+        opt_item = self.graph_item.optimizers[0] if self.graph_item.optimizers else None
+        opt = opt_item.optimizer if opt_item else None
+        if opt is None:
+            return
+        for gidx, group in enumerate(opt.param_groups):
+            cur = self._group_hyper.get(gidx)
+            if cur is None:
+                continue
+            for k, v in group.items():
+                if k != "params" and cur.get(k) != v:
+                    cur[k] = v
+
+    def set_batch_fraction(self, fraction: Optional[float]):
+        """Per-rank batch fraction n_r/N for the current step's gradients.
+        Used as the gradient weight so uneven feed splits produce the exact
+        weighted average (reference tests/integration/cases/c0.py:92-119)."""
+        self._batch_fraction = fraction
+
+    def grad_scale(self) -> float:
+        """Weight applied to this rank's gradients before the summing
+        collective: batch fraction if known, else 1/world."""
+        if self._batch_fraction is not None:
+            return float(self._batch_fraction)
+        return 1.0 / self.world_size
 
     def _build_plans(self):
         vars_by_name = self.graph_item.trainable_var_op_to_var
@@ -248,12 +298,18 @@ class DistributedEngine:
             item = vars_by_name.get(node.var_name)
             if item is None or item.param is None:
                 continue
-            cls_name, hyper = self._hyper_for(item.param)
+            cls_name, hyper, gidx = self._hyper_for(item.param)
             plan = VarPlan(name=node.var_name, param=item.param,
-                           sparse=item.is_sparse, cls_name=cls_name, hyper=hyper)
+                           sparse=item.is_sparse, cls_name=cls_name,
+                           hyper=hyper, group_index=gidx)
             if node.part_config and not getattr(
                     item.param, "_autodist_shard_local", False):
                 slices = make_shard_slices(item.shape, node.partitioner)
+                if len(slices) != len(node.part_config):
+                    raise ValueError(
+                        f"variable {node.var_name}: partitioner produced "
+                        f"{len(slices)} shards but strategy has "
+                        f"{len(node.part_config)} part_config entries")
                 for sl, part in zip(slices, node.part_config):
                     plan.shards.append(self._make_shard(part, sl))
             else:
@@ -323,9 +379,13 @@ class DistributedEngine:
                 continue
             if whole and sh0.kind == "allreduce":
                 plan.bucketed = True
+                # group_index in the key keeps buckets homogeneous in LIVE
+                # param_group (two groups with equal initial hyper may
+                # diverge later under a scheduler)
                 bucket_items.append((plan.param, sh0.group, sh0.compressor,
                                      plan.cls_name, plan.hyper,
-                                     _hyper_key(plan.hyper)))
+                                     (plan.group_index,
+                                      _hyper_key(plan.hyper))))
             elif sh0.kind == "allreduce":
                 # partitioned AR: per-shard direct reducers
                 for sh in plan.shards:
@@ -348,6 +408,21 @@ class DistributedEngine:
                     self._ps_outstanding[sh.name + f"/{sh.slice.start if sh.slice else 0}"] = []
         # PS/partitioned params keep ordinary grads; bucketed params get views
         self.buckets = build_buckets(bucket_items, self.device, self.bucket_bytes)
+        # deterministic cross-rank collective keys (reference
+        # collective_key.py:43-70): instance key = md5 of member var names,
+        # group key = the replica device set; engine.step() flushes unissued
+        # buckets in instance-key order so every rank enqueues RCCL
+        # collectives identically regardless of hook timing
+        from autodist_amd.parallel.collective_key import get_collective_keys
+        ck = get_collective_keys()
+        name_of = {id(pl.param): pl.name for pl in self.var_plans}
+        devices = self.strategy.graph_config.replicas or [
+            f"127.0.0.1:GPU:{r}" for r in range(self.world_size)]
+        for b in self.buckets:
+            members = ",".join(sorted(name_of.get(id(p), str(id(p)))
+                                      for p in b.params))
+            b.instance_key = ck.generate_instance_key(members)
+            b.group_key = ck.generate_group_key(devices)
         param_to_bucket = {}
         for b in self.buckets:
             for p in b.params:
@@ -408,16 +483,27 @@ class DistributedEngine:
             return
         engine = self
 
-        def step(closure=None):  # noqa: ARG001
-            engine.step()
+        import types
 
-        def zero_grad(set_to_none=True):  # noqa: ARG001
+        def step(_opt, closure=None):
+            loss = None
+            if closure is not None:
+                # torch.optim contract: re-evaluate the model under grad
+                # and return the loss (LBFGS-style loops)
+                with torch.enable_grad():
+                    loss = closure()
+            engine.step()
+            return loss
+
+        def zero_grad(_opt, set_to_none=True):  # noqa: ARG001
             engine.zero_grad()
 
+        # bound methods (not bare functions): torch LR schedulers wrap
+        # opt.step and require __func__/__self__ on it
         opt._autodist_orig_step = opt.step
-        opt.step = step
+        opt.step = types.MethodType(step, opt)
         opt._autodist_orig_zero_grad = opt.zero_grad
-        opt.zero_grad = zero_grad
+        opt.zero_grad = types.MethodType(zero_grad, opt)
 
     # ------------------------------------------------------------------ steps
     def zero_grad(self):
@@ -433,8 +519,12 @@ class DistributedEngine:
 
     def step(self):
         """Synchronize gradients + apply updates. Call after backward."""
-        # 1) flush collectives not yet issued (e.g. params w/o grads)
-        for b in self.buckets:
+        # 0) pick up live LR-scheduler / param_groups changes
+        self._refresh_hyper()
+        # 1) flush collectives not yet issued (e.g. params w/o grads) in
+        #    instance-key order — deterministic across ranks regardless of
+        #    hook timing (reference collective_key.py:60-70 semantics)
+        for b in sorted(self.buckets, key=lambda b: b.instance_key):
             if not b._issued:
                 b.issue(self)
         for plan in self.var_plans:
@@ -473,6 +563,15 @@ class DistributedEngine:
             for sh in plan.shards:
                 if sh.kind == "ps":
                     self._consume_ps_rounds(plan, sh)
+        # 7) reset per-step issue state HERE (not only in zero_grad): user
+        #    code may call model.zero_grad(set_to_none=True) instead of the
+        #    patched optimizer.zero_grad, which would leave _issued stuck
+        for b in self.buckets:
+            b.reset()
+        for plan in self.var_plans:
+            for sh in plan.shards:
+                if sh.reducer is not None:
+                    sh.reducer.reset()
         self._step_count += 1
 
     # -- dense AR apply ----------------------------------------------------
@@ -558,11 +657,13 @@ class DistributedEngine:
             # dense grad on a sparse-flagged var: treat all rows as touched
             indices = torch.arange(grad.shape[0], device=grad.device)
             values = grad
+        # weight BEFORE the gather so uneven batch splits produce the exact
+        # weighted average after cross-rank coalesce (c0.py:92-119 semantics)
+        values = values * self.grad_scale()
         if self.world_size > 1:
             indices, values = allgather_sparse(
                 indices, values, self.world_size, self.process_group)
         indices, values = coalesce_rows(indices, values)
-        values = values / float(self.world_size)
         # replicated rowwise apply: identical on every rank == PS result
         sh = plan.shards[0]
         if sh.state is None:

@@ -54,7 +54,7 @@ class PSSynchronizer:
         gbuf = gview if gview.is_contiguous() else gview.contiguous()
 
         def round_body():
-            gbuf.mul_(1.0 / engine.world_size)
+            gbuf.mul_(engine.grad_scale())
             dist.reduce(gbuf, dst=sh.owner_rank, group=engine.process_group)
             if engine.rank == sh.owner_rank:
                 apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
@@ -87,7 +87,7 @@ class PSSynchronizer:
             # CPU/gloo: async handles so non-owner workers RUN AHEAD within
             # the staleness bound (the c9-verified behavior) — only the owner
             # blocks, on the reduced gradient it must apply.
-            gbuf.mul_(1.0 / engine.world_size)
+            gbuf.mul_(engine.grad_scale())
             h_red = dist.reduce(gbuf, dst=sh.owner_rank,
                                 group=engine.process_group, async_op=True)
             if engine.rank == sh.owner_rank:

@@ -26,6 +26,11 @@ class Remapper:
         self.world_size = world_size
         self.device = device
         self.process_group = process_group
+        # n_local/N of the last feed split — np.array_split produces UNEVEN
+        # shards when N % world != 0; gradients and scalar fetches must then
+        # be weighted by batch fraction, not 1/world (the reference asserts
+        # the weighted average: tests/integration/cases/c0.py:92-119)
+        self.batch_fraction: float = 1.0 / max(world_size, 1)
 
     # -- feeds -------------------------------------------------------------
     def remap_feed(self, value: Any) -> Any:
@@ -33,12 +38,18 @@ class Remapper:
         (reference _remap_feed, remapper.py:81-123)."""
         if isinstance(value, np.ndarray):
             if self.world_size > 1 and value.ndim > 0:
+                total = value.shape[0]
                 value = np.array_split(value, self.world_size)[self.rank]
+                if total:
+                    self.batch_fraction = value.shape[0] / total
             t = torch.from_numpy(np.ascontiguousarray(value))
             return t.to(self.device)
         if isinstance(value, torch.Tensor):
             if self.world_size > 1 and value.dim() > 0:
+                total = value.shape[0]
                 value = torch.tensor_split(value, self.world_size)[self.rank]
+                if total:
+                    self.batch_fraction = value.shape[0] / total
             return value.to(self.device)
         return value  # scalars / python objects duplicated
 
@@ -55,9 +66,12 @@ class Remapper:
         if self.world_size <= 1:
             return value
         if value.dim() == 0:
-            out = value.clone()
+            # batch-fraction-weighted mean: with an even split this is the
+            # plain mean; with uneven np.array_split shards it matches the
+            # global-batch scalar (e.g. the loss over the whole batch)
+            out = value.clone() * self.batch_fraction
             dist.all_reduce(out, op=dist.ReduceOp.SUM, group=self.process_group)
-            return out / self.world_size
+            return out
         parts = allgatherv(value.contiguous(), self.world_size,
                            self.process_group)
         return torch.cat(parts, dim=0)

@@ -55,6 +55,9 @@ class WrappedSession:
         if self._check_mutation:
             self._assert_not_mutated()
         feeds = self.remapper.remap_feed_dict(feed_dict)
+        # propagate the (possibly uneven) split fraction so gradients are
+        # weighted by per-rank batch size (reference c0.py:92-119)
+        self.engine.set_batch_fraction(self.remapper.batch_fraction)
         trace = bool(options and options.get("trace"))
         if trace:
             with torch.profiler.profile(
