@@ -1,10 +1,18 @@
-"""Flagship benchmark: ResNet-50 data-parallel training, images/sec.
+"""Driver benchmark for the BASELINE configs.
 
-BASELINE.json metric: "images/sec ResNet-50 AllReduce at 1/2/4/8 MI355X;
-scaling efficiency". bf16 autocast compute, fp32 params, synthetic
-ImageNet-shape data, random-init weights, AllReduce strategy over RCCL/xGMI.
+Default (no flags): ResNet-50 data-parallel training, images/sec — the
+headline metric in BASELINE.json ("images/sec ResNet-50 AllReduce at
+1/2/4/8 MI355X; scaling efficiency"). bf16 autocast compute, fp32 params,
+synthetic data, random-init weights, AllReduce strategy over RCCL/xGMI.
 
-Usage:  python bench.py --gpus N --steps K --warmup W
+--model selects the other BASELINE configs (each emits the same JSON
+contract, driver-verifiable):
+  bert   — BERT-base Parallax (config #3), sequences/sec
+  ncf    — NCF PartitionedPS, sparse embedding push/pull (config #4),
+           samples/sec
+  lm1b   — LM1B LSTM, simulator-selected strategy (config #5), words/sec
+
+Usage:  python bench.py [--model M] --gpus N --steps K --warmup W
 The driver launches N>1 via torch.distributed.run (one rank per GPU); run
 standalone and it re-execs itself under the launcher.
 """
@@ -24,18 +32,21 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch-size", type=int, default=512,
-                   help="per-GPU batch (weak scaling; 288 GB HBM3E/GPU)")
-    p.add_argument("--model", default="resnet50")
-    p.add_argument("--strategy", default="AllReduce")
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "resnet101", "bert", "ncf", "lm1b"])
+    p.add_argument("--batch-size", type=int, default=None,
+                   help="per-GPU batch (weak scaling; default per model)")
+    p.add_argument("--strategy", default=None,
+                   help="strategy builder name (default per model)")
     p.add_argument("--bucket-mb", type=int, default=25)
+    p.add_argument("--seq-len", type=int, default=None)
     p.add_argument("--image-size", type=int, default=224)
     p.add_argument("--no-channels-last", action="store_true")
     p.add_argument("--no-fused-bn", action="store_true",
                    help="disable the hand-written gfx950 fused BN kernels")
     p.add_argument("--hipgraph", choices=["auto", "on", "off"], default="auto",
                    help="capture the train step in a hipGraph (auto: on for "
-                        "world_size==1)")
+                        "graph-safe models)")
     return p.parse_args()
 
 
@@ -51,6 +62,164 @@ def relaunch_under_torchrun(args):
            os.path.abspath(__file__)] + sys.argv[1:]
     os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     os.execvpe(sys.executable, cmd, os.environ)
+
+
+def _make_engine(model, opt, strategy_name, rank, world, device, bucket_mb):
+    from autodist_amd import strategy as strat
+    from autodist_amd.graph_item import GraphItem
+    from autodist_amd.parallel.engine import DistributedEngine
+    from autodist_amd.resource_spec import ResourceSpec
+
+    g = GraphItem()
+    g.extend_model(model)
+    g.extend_optimizer_info(opt)
+    builder = getattr(strat, strategy_name)()
+    strategy = builder.build(g, ResourceSpec())
+    if len(strategy.graph_config.replicas) != world:
+        strategy.graph_config.replicas = [
+            f"127.0.0.1:GPU:{r}" for r in range(world)]
+    engine = DistributedEngine(g, strategy, rank=rank, world_size=world,
+                               device=device,
+                               bucket_bytes=bucket_mb * 1024 * 1024)
+    engine.setup()
+    return engine
+
+
+def build_resnet(args, device, rank, world, use_cuda):
+    from autodist_amd.models import resnet
+    strategy_name = args.strategy or "AllReduce"
+    B = args.batch_size or 512
+    fused = use_cuda and not args.no_fused_bn
+    model = getattr(resnet, args.model)(num_classes=1000, fused=fused)
+    model = model.to(device)
+    channels_last = use_cuda and not args.no_channels_last
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+    engine = _make_engine(model, opt, strategy_name, rank, world, device,
+                          args.bucket_mb)
+    x = torch.randn(B, 3, args.image_size, args.image_size, device=device)
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (B,), device=device)
+    loss_fn = torch.nn.CrossEntropyLoss()
+
+    def step():
+        opt.zero_grad()
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_cuda):
+            loss = loss_fn(model(x), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    return dict(step=step, engine=engine, items_per_step=B,
+                metric=f"images/sec ResNet-50 {strategy_name}",
+                unit="images/sec", graph_safe=True,
+                config={"model": args.model, "global_batch": world * B,
+                        "seq_len": None, "parallelism": f"dp{world}",
+                        "strategy": strategy_name,
+                        "image_size": args.image_size})
+
+
+def build_bert(args, device, rank, world, use_cuda):
+    """BASELINE config #3: BERT-base Parallax (hybrid PS + AllReduce).
+    Reference workload: examples/benchmark/bert.py (TF-model-garden)."""
+    from autodist_amd.models import bert
+    strategy_name = args.strategy or "Parallax"
+    B = args.batch_size or 32
+    S = args.seq_len or 128
+    model = bert.bert_base().to(device)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-4, weight_decay=0.01)
+    engine = _make_engine(model, opt, strategy_name, rank, world, device,
+                          args.bucket_mb)
+    vocab = model.bert.cfg.vocab_size
+    ids = torch.randint(0, vocab, (B, S), device=device)
+    labels = ids.clone()
+    labels[:, ::2] = -100  # predict every other position (synthetic MLM)
+    nsp = torch.randint(0, 2, (B,), device=device)
+
+    def step():
+        opt.zero_grad()
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_cuda):
+            loss = model.loss(ids, labels, nsp)
+        loss.backward()
+        opt.step()
+        return loss
+
+    return dict(step=step, engine=engine, items_per_step=B,
+                metric=f"sequences/sec BERT-base {strategy_name}",
+                unit="sequences/sec", graph_safe=True,
+                config={"model": "bert_base", "global_batch": world * B,
+                        "seq_len": S, "parallelism": f"dp{world}",
+                        "strategy": strategy_name})
+
+
+def build_ncf(args, device, rank, world, use_cuda):
+    """BASELINE config #4: NCF PartitionedPS with sparse-embedding
+    push/pull. Reference workload: examples/benchmark/ncf.py."""
+    from autodist_amd.models.ncf import ncf_movielens
+    strategy_name = args.strategy or "PartitionedPS"
+    B = args.batch_size or 4096
+    model = ncf_movielens(sparse=True).to(device)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+    engine = _make_engine(model, opt, strategy_name, rank, world, device,
+                          args.bucket_mb)
+    users = torch.randint(0, 138493, (B,), device=device)
+    items = torch.randint(0, 26744, (B,), device=device)
+    labels = torch.randint(0, 2, (B,), device=device)
+
+    def step():
+        opt.zero_grad()
+        loss = model.loss(users, items, labels)
+        loss.backward()
+        opt.step()
+        return loss
+
+    return dict(step=step, engine=engine, items_per_step=B,
+                metric=f"samples/sec NCF {strategy_name}",
+                unit="samples/sec", graph_safe=False,  # sparse/PS host logic
+                config={"model": "ncf_movielens", "global_batch": world * B,
+                        "seq_len": None, "parallelism": f"dp{world}",
+                        "strategy": strategy_name})
+
+
+def build_lm1b(args, device, rank, world, use_cuda):
+    """BASELINE config #5: LM1B LSTM LM with the simulator-selected
+    strategy. Reference workload: examples/lm1b/lm1b_train.py."""
+    from autodist_amd.models.lm1b import lm1b_full
+    strategy_name = args.strategy or "AutoStrategy"
+    B = args.batch_size or 128
+    S = args.seq_len or 20
+    model = lm1b_full().to(device)
+    opt = torch.optim.Adagrad(model.parameters(), lr=0.2)
+    engine = _make_engine(model, opt, strategy_name, rank, world, device,
+                          args.bucket_mb)
+    vocab = model.emb.num_embeddings
+    tokens = torch.randint(0, vocab, (B, S), device=device)
+    targets = torch.randint(0, vocab, (B, S), device=device)
+
+    def step():
+        opt.zero_grad()
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_cuda):
+            loss = model.loss(tokens, targets)
+        loss.backward()
+        opt.step()
+        return loss
+
+    return dict(step=step, engine=engine, items_per_step=B * S,
+                metric=f"words/sec LM1B {strategy_name}",
+                unit="words/sec", graph_safe=False,  # cuDNN-style LSTM
+                config={"model": "lm1b_lstm", "global_batch": world * B,
+                        "seq_len": S, "parallelism": f"dp{world}",
+                        "strategy": strategy_name, "vocab": vocab})
+
+
+BUILDERS = {"resnet50": build_resnet, "resnet101": build_resnet,
+            "bert": build_bert, "ncf": build_ncf, "lm1b": build_lm1b}
 
 
 def main():
@@ -72,55 +241,15 @@ def main():
         torch.backends.cudnn.benchmark = True
         torch.backends.cuda.matmul.allow_tf32 = False
 
-    from autodist_amd import strategy as strat
-    from autodist_amd.graph_item import GraphItem
-    from autodist_amd.models import resnet
-    from autodist_amd.parallel.engine import DistributedEngine
-    from autodist_amd.resource_spec import ResourceSpec
-
     torch.manual_seed(1234)
-    fused = use_cuda and not args.no_fused_bn
-    model = getattr(resnet, args.model)(num_classes=1000, fused=fused)
-    model = model.to(device)
-    channels_last = use_cuda and not args.no_channels_last
-    if channels_last:
-        model = model.to(memory_format=torch.channels_last)
-
-    g = GraphItem()
-    g.extend_model(model)
-    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
-                          weight_decay=1e-4)
-    g.extend_optimizer_info(opt)
-    builder = getattr(strat, args.strategy)()
-    rs = ResourceSpec()
-    strategy = builder.build(g, rs)
-    if len(strategy.graph_config.replicas) != world:
-        strategy.graph_config.replicas = [
-            f"127.0.0.1:GPU:{r}" for r in range(world)]
-    engine = DistributedEngine(g, strategy, rank=rank, world_size=world,
-                               device=device,
-                               bucket_bytes=args.bucket_mb * 1024 * 1024)
-    engine.setup()
-
-    B = args.batch_size
-    x = torch.randn(B, 3, args.image_size, args.image_size, device=device)
-    if channels_last:
-        x = x.contiguous(memory_format=torch.channels_last)
-    y = torch.randint(0, 1000, (B,), device=device)
-    loss_fn = torch.nn.CrossEntropyLoss()
-
-    amp_dtype = torch.bfloat16
-    amp_enabled = use_cuda
-
-    def step():
-        opt.zero_grad()
-        with torch.autocast(device_type="cuda", dtype=amp_dtype,
-                            enabled=amp_enabled):
-            out = model(x)
-            loss = loss_fn(out, y)
-        loss.backward()
-        opt.step()
-        return loss
+    wl = BUILDERS[args.model](args, device, rank, world, use_cuda)
+    step, engine = wl["step"], wl["engine"]
+    # PS rounds and sparse sync involve host-side queue state that a graph
+    # replay would freeze — never capture those
+    has_host_sync = any(sh.kind == "ps" for p in engine.var_plans
+                        for sh in p.shards) \
+        or any(p.sparse for p in engine.var_plans)
+    wl["graph_safe"] = wl["graph_safe"] and not has_host_sync
 
     import torch.distributed as dist
 
@@ -130,11 +259,13 @@ def main():
         if use_cuda:
             torch.cuda.synchronize(device)
 
-    # hipGraph capture: the per-step kernel chain (53 fused-BN trios + convs +
-    # the fused optimizer launch) is launch-bound in eager mode; replaying a
-    # captured graph removes host launch + python overhead entirely.
+    # hipGraph capture: the per-step kernel chain is launch-bound in eager
+    # mode; replaying a captured graph removes host launch + python overhead
+    # entirely. world>1: RCCL collectives are capturable (torch NCCL graph
+    # support); non-graph-safe models (sparse/PS host-side logic, cuDNN RNN)
+    # stay eager unless forced.
     use_graph = use_cuda and (args.hipgraph == "on" or
-                              (args.hipgraph == "auto" and world == 1))
+                              (args.hipgraph == "auto" and wl["graph_safe"]))
     run_step = step
     for _ in range(args.warmup):
         step()
@@ -146,6 +277,8 @@ def main():
                 for _ in range(2):
                     step()
             torch.cuda.current_stream().wait_stream(s)
+            if world > 1:
+                barrier_sync()  # all ranks enter capture together
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 static_loss = step()
@@ -175,11 +308,11 @@ def main():
     engine.drain()
 
     if rank == 0:
-        ips = world * B * args.steps / dt
+        value = world * wl["items_per_step"] * args.steps / dt
         result = {
-            "metric": "images/sec ResNet-50 AllReduce",
-            "value": round(ips, 2),
-            "unit": "images/sec",
+            "metric": wl["metric"],
+            "value": round(value, 2),
+            "unit": wl["unit"],
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -187,12 +320,9 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if use_cuda else "fp32",
             "data": "synthetic",
-            "config": {"model": args.model, "global_batch": world * B,
-                       "seq_len": None, "parallelism": f"dp{world}",
-                       "strategy": args.strategy,
-                       "image_size": args.image_size},
+            "config": wl["config"],
         }
         print(json.dumps(result))
 
