@@ -230,7 +230,11 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     use_cuda = torch.cuda.is_available()
-    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) \
+    # modulo device count: lets an N-rank job run on fewer GPUs (e.g. the
+    # RCCL-validation mode: 2 ranks sharing the single leased MI355X)
+    device = torch.device(
+        "cuda",
+        int(os.environ.get("LOCAL_RANK", 0)) % torch.cuda.device_count()) \
         if use_cuda else torch.device("cpu")
     if use_cuda:
         torch.cuda.set_device(device)

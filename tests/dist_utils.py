@@ -1,8 +1,11 @@
-"""Multi-process test harness: gloo world_size>1 on localhost.
+"""Multi-process test harness: world_size>1 on localhost.
 
 Mirrors the reference's per-test process isolation (tests/integration/
 test_all.py:55-70 runs each case in a forked Process); here each case runs
-in spawned workers joined by a gloo file-store rendezvous.
+in spawned workers joined by a file-store rendezvous. Backend "gloo" for
+CPU-only CI; backend "nccl" (= RCCL on ROCm) for GPU validation — with
+fewer GPUs than ranks, ranks share devices modulo the device count (the
+2-ranks-on-1-MI355X RCCL validation mode).
 """
 import os
 import tempfile
@@ -12,12 +15,16 @@ import torch.distributed as dist
 import torch.multiprocessing as mp
 
 
-def _worker(rank, world_size, init_file, fn, args, err_queue):
+def _worker(rank, world_size, init_file, fn, args, err_queue, backend):
     try:
         os.environ["RANK"] = str(rank)
         os.environ["WORLD_SIZE"] = str(world_size)
+        if backend == "nccl":
+            import torch
+            os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+            torch.cuda.set_device(rank % torch.cuda.device_count())
         dist.init_process_group(
-            "gloo", init_method=f"file://{init_file}",
+            backend, init_method=f"file://{init_file}",
             rank=rank, world_size=world_size)
         fn(rank, world_size, *args)
         dist.barrier()
@@ -27,7 +34,7 @@ def _worker(rank, world_size, init_file, fn, args, err_queue):
         raise
 
 
-def run_distributed(fn, world_size=2, args=(), timeout=300):
+def run_distributed(fn, world_size=2, args=(), timeout=300, backend="gloo"):
     """Run fn(rank, world_size, *args) in `world_size` spawned processes."""
     with tempfile.TemporaryDirectory() as td:
         init_file = os.path.join(td, "rendezvous")
@@ -37,7 +44,7 @@ def run_distributed(fn, world_size=2, args=(), timeout=300):
         for rank in range(world_size):
             p = ctx.Process(target=_worker,
                             args=(rank, world_size, init_file, fn, args,
-                                  err_queue))
+                                  err_queue, backend))
             p.start()
             procs.append(p)
         failed = []
