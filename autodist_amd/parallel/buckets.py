@@ -114,7 +114,7 @@ class Bucket:
     def issue(self, engine):
         self._issued = True
         self.ensure_views()
-        if engine.world_size <= 1:
+        if not engine.collectives_active:
             return
         if engine.device.type == "cuda":
             ev = torch.cuda.Event()
@@ -134,7 +134,7 @@ class Bucket:
 
     def finalize(self, engine):
         """Make the compute stream depend on this bucket's reduced result."""
-        if engine.world_size <= 1 or not self._issued:
+        if not engine.collectives_active or not self._issued:
             return
         if engine.device.type == "cuda":
             with torch.cuda.stream(engine.comm_stream):

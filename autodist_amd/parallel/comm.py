@@ -12,13 +12,15 @@ import torch
 import torch.distributed as dist
 
 
-def allgatherv(tensor: torch.Tensor, world_size: int, group=None
-               ) -> List[torch.Tensor]:
+def allgatherv(tensor: torch.Tensor, world_size: int, group=None,
+               force: bool = False) -> List[torch.Tensor]:
     """All-gather tensors whose dim-0 length differs per rank.
 
     Returns the per-rank tensors (views of one padded buffer, trimmed).
+    `force` executes the real collectives even at world_size==1 (the 1-GPU
+    RCCL validation mode, AUTODIST_FORCE_COLLECTIVES).
     """
-    if world_size <= 1:
+    if world_size <= 1 and not force:
         return [tensor]
     n_local = torch.tensor([tensor.shape[0]], dtype=torch.int64,
                            device=tensor.device)
@@ -40,12 +42,12 @@ def allgatherv(tensor: torch.Tensor, world_size: int, group=None
 
 
 def allgather_sparse(indices: torch.Tensor, values: torch.Tensor,
-                     world_size: int, group=None
+                     world_size: int, group=None, force: bool = False
                      ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Gather a row-sparse gradient (indices [nnz], values [nnz, dim...])
     from all ranks; returns concatenated (indices, values)."""
-    idx_parts = allgatherv(indices, world_size, group)
-    val_parts = allgatherv(values, world_size, group)
+    idx_parts = allgatherv(indices, world_size, group, force=force)
+    val_parts = allgatherv(values, world_size, group, force=force)
     return torch.cat(idx_parts), torch.cat(val_parts)
 
 

@@ -113,7 +113,7 @@ class ShardReducer:
         if self.param.grad is None:
             return
         self._issued = True
-        if engine.world_size <= 1:
+        if not engine.collectives_active:
             return
         if engine.device.type == "cuda":
             ev = torch.cuda.Event()
@@ -132,7 +132,7 @@ class ShardReducer:
         self._reduced_tensor = t
 
     def finalize(self, engine):
-        if engine.world_size <= 1 or not self._issued:
+        if not engine.collectives_active or not self._issued:
             return
         if engine.device.type == "cuda":
             with torch.cuda.stream(engine.comm_stream):
@@ -189,13 +189,29 @@ class DistributedEngine:
         # per-rank gradient weight for uneven batch splits: weighted average
         # sum_r (n_r / N) * g_r (reference c0.py:92-119). None => 1/world.
         self._batch_fraction: Optional[float] = None
+        # AUTODIST_FORCE_COLLECTIVES=1: execute every RCCL collective even at
+        # world_size==1 (sum over one rank == identity, so numerics are
+        # unchanged). This is the 1-GPU hardware-validation mode: real
+        # ncclAllReduce/Broadcast/Reduce/AllGather enqueue, comm-stream event
+        # ordering, compressor wire handles and hipGraph x RCCL capture all
+        # run exactly as they do at world 8 — RCCL itself refuses >1 rank
+        # per device ("Duplicate GPU detected"), so this is the deepest
+        # single-GPU proof available.
+        self._force_collectives = os.environ.get(
+            "AUTODIST_FORCE_COLLECTIVES", "") in ("1", "True")
         # note: RCCL supports ReduceOp.AVG, but the mean is instead fused as
         # a scale into the compress/cast kernels (one code path for gloo +
         # every compressor)
 
+    @property
+    def collectives_active(self) -> bool:
+        """True when the engine issues real collectives (world>1, or the
+        world-1 RCCL validation mode AUTODIST_FORCE_COLLECTIVES)."""
+        return self.world_size > 1 or self._force_collectives
+
     # ------------------------------------------------------------------ set-up
     def maybe_init_process_group(self):
-        if self.world_size <= 1 or dist.is_initialized():
+        if not self.collectives_active or dist.is_initialized():
             return
         backend = "nccl" if self.device.type == "cuda" else "gloo"
         if self.device.type == "cuda":
@@ -357,7 +373,7 @@ class DistributedEngine:
         return ShardPlan(name=node.var_name, kind="allreduce", slice=sl)
 
     def _sync_initial_params(self):
-        if self.world_size <= 1:
+        if not self.collectives_active:
             return
         for plan in self.var_plans:
             if getattr(plan.param, "_autodist_shard_local", False):
@@ -660,9 +676,10 @@ class DistributedEngine:
         # weight BEFORE the gather so uneven batch splits produce the exact
         # weighted average after cross-rank coalesce (c0.py:92-119 semantics)
         values = values * self.grad_scale()
-        if self.world_size > 1:
+        if self.collectives_active:
             indices, values = allgather_sparse(
-                indices, values, self.world_size, self.process_group)
+                indices, values, self.world_size, self.process_group,
+                force=self._force_collectives)
         indices, values = coalesce_rows(indices, values)
         # replicated rowwise apply: identical on every rank == PS result
         sh = plan.shards[0]

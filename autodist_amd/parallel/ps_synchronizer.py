@@ -68,7 +68,7 @@ class PSSynchronizer:
                 sh.stage.copy_(sh.master)
 
         key = engine._ps_key(sh)
-        if engine.world_size <= 1:
+        if not engine.collectives_active:
             apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
                                   [sh.state], plan.hyper)
             sh.stage.copy_(sh.master)

@@ -47,6 +47,9 @@ def parse_args():
     p.add_argument("--hipgraph", choices=["auto", "on", "off"], default="auto",
                    help="capture the train step in a hipGraph (auto: on for "
                         "graph-safe models)")
+    p.add_argument("--force-collectives", action="store_true",
+                   help="execute real RCCL collectives even at world 1 "
+                        "(1-GPU hardware validation of the comm path)")
     return p.parse_args()
 
 
@@ -224,6 +227,8 @@ BUILDERS = {"resnet50": build_resnet, "resnet101": build_resnet,
 
 def main():
     args = parse_args()
+    if args.force_collectives:
+        os.environ["AUTODIST_FORCE_COLLECTIVES"] = "1"
     if args.gpus > 1 and "RANK" not in os.environ:
         relaunch_under_torchrun(args)
 

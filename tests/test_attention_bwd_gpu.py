@@ -1,19 +1,14 @@
 """MFMA attention BACKWARD vs fp32 autograd reference.
 
-DRAFT-gated: the backward kernels are compile-checked but not yet
-GPU-validated (round-1 GPU budget was exhausted after the forward landed);
-set AUTODIST_EXPERIMENTAL=1 to run — this is round 2's first GPU task."""
+GPU-validated in round 2 (both cases passed on MI355X, 2026-09-14) — the
+kernels are now the default training path for unmasked bf16 D=64 attention
+via ops/fused_attention.FusedAttentionFn."""
 import math
-import os
 
 import pytest
 import torch
 
-pytestmark = [
-    pytest.mark.gpu,
-    pytest.mark.skipif(os.environ.get("AUTODIST_EXPERIMENTAL") != "1",
-                       reason="draft kernel: set AUTODIST_EXPERIMENTAL=1"),
-]
+pytestmark = pytest.mark.gpu
 
 
 @pytest.mark.parametrize("B,H,S", [(1, 1, 32), (2, 3, 128)])
@@ -42,3 +37,28 @@ def test_attn_bwd_matches_autograd(B, H, S):
                            ("dv", dv, vf.grad)):
         err = (got.float() - ref).abs().max().item()
         assert err < tol, f"{name} max err {err}"
+
+
+def test_fused_attention_autograd_path():
+    """FusedAttentionFn end-to-end: loss.backward() through the fused
+    kernels matches the SDPA training path."""
+    from autodist_amd.ops.fused_attention import fused_sdpa
+    torch.manual_seed(2)
+    B, H, S, D = 2, 4, 64, 64
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    out = fused_sdpa(q, k, v)
+    assert out.requires_grad and out.grad_fn is not None
+    loss = (out.float() ** 2).mean()
+    loss.backward()
+    # reference grads via SDPA on fp32 copies
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    of = torch.nn.functional.scaled_dot_product_attention(
+        qf, kf, vf, scale=1.0 / math.sqrt(D))
+    (of ** 2).mean().backward()
+    for g, r in ((q.grad, qf.grad), (k.grad, kf.grad), (v.grad, vf.grad)):
+        assert (g.float() - r).abs().max().item() < 6e-2
