@@ -176,7 +176,7 @@ class DistributedEngine:
         self.buckets: List[Bucket] = []
         self.var_plans: List[VarPlan] = []
         self._hook_handles = []
-        self._ps_outstanding: Dict[str, list] = {}
+        self.ps_groups: list = []
         self._replica_ranks: Dict[str, int] = {}
         self._step_count = 0
         self._fallback_user_opt = False
@@ -382,6 +382,7 @@ class DistributedEngine:
 
     def _build_buckets_and_hooks(self):
         bucket_items = []
+        ps_items = []
         hooked = set()
         for plan in self.var_plans:
             if plan.sparse:
@@ -412,16 +413,18 @@ class DistributedEngine:
                         plan.param.register_post_accumulate_grad_hook(
                             self._make_shard_hook(plan)))
             else:
-                # PS shards: allocate master/stage on owner/all ranks
-                for sh in plan.shards:
-                    view = sh.slice.view(plan.param.data) if sh.slice \
-                        else plan.param.data
-                    sh.stage = view.detach().clone().contiguous()
-                    if self.rank == sh.owner_rank:
-                        sh.master = view.detach().clone().contiguous()
-                        sh.state = apply_mod.make_state(
-                            plan.cls_name, sh.master, plan.hyper)
-                    self._ps_outstanding[sh.name + f"/{sh.slice.start if sh.slice else 0}"] = []
+                # PS shards: coalesced into per-owner flat groups below
+                ps_items.extend((plan, sh) for sh in plan.shards)
+                if id(plan.param) not in hooked:
+                    hooked.add(id(plan.param))
+                    self._hook_handles.append(
+                        plan.param.register_post_accumulate_grad_hook(
+                            self._make_ps_hook(plan)))
+        # one reduce+broadcast per OWNER per step, hook-issued so PS traffic
+        # overlaps backward (reference dataflow overlap,
+        # ps_synchronizer.py:250-332; round-1 issued per-shard pairs)
+        from autodist_amd.parallel.ps_synchronizer import PSSynchronizer
+        self.ps_groups = PSSynchronizer.build_groups(self, ps_items)
         # PS/partitioned params keep ordinary grads; bucketed params get views
         self.buckets = build_buckets(bucket_items, self.device, self.bucket_bytes)
         # deterministic cross-rank collective keys (reference
@@ -472,6 +475,18 @@ class DistributedEngine:
                 return
             for sh in plan.shards:
                 sh.reducer.issue(self)
+        return hook
+
+    def _make_ps_hook(self, plan: VarPlan):
+        def hook(_param):
+            if self._accumulating:
+                return
+            seen = set()  # a plan's shards may live in different owner groups
+            for sh in plan.shards:
+                grp = getattr(sh, "owner_group", None)
+                if grp is not None and id(grp) not in seen:
+                    seen.add(id(grp))
+                    grp.mark_param_ready(self, plan)
         return hook
 
     def no_sync(self):
@@ -558,13 +573,10 @@ class DistributedEngine:
         for plan in self.var_plans:
             if plan.sparse:
                 self._sync_and_apply_sparse(plan)
-        # 4) PS rounds
-        for plan in self.var_plans:
-            if plan.sparse:
-                continue
-            for sh in plan.shards:
-                if sh.kind == "ps":
-                    self._issue_ps_round(plan, sh)
+        # 4) PS rounds: one reduce+apply+broadcast per owner group (those
+        #    already hook-issued just complete; the rest flush here)
+        for grp in self.ps_groups:
+            grp.apply_and_broadcast(self)
         # 5) dense applies (AR vars) — grouped multi-tensor
         if self._fallback_user_opt:
             opt = self.graph_item.optimizer
@@ -573,17 +585,15 @@ class DistributedEngine:
         else:
             self._apply_dense_updates()
         # 6) consume due PS rounds (staleness bound)
-        for plan in self.var_plans:
-            if plan.sparse:
-                continue
-            for sh in plan.shards:
-                if sh.kind == "ps":
-                    self._consume_ps_rounds(plan, sh)
+        for grp in self.ps_groups:
+            grp.consume_due(self)
         # 7) reset per-step issue state HERE (not only in zero_grad): user
         #    code may call model.zero_grad(set_to_none=True) instead of the
         #    patched optimizer.zero_grad, which would leave _issued stuck
         for b in self.buckets:
             b.reset()
+        for grp in self.ps_groups:
+            grp.reset()
         for plan in self.var_plans:
             for sh in plan.shards:
                 if sh.reducer is not None:
@@ -623,41 +633,10 @@ class DistributedEngine:
             states = [x[2] for x in items]
             apply_mod.apply_dense(cls_name, params, grads, states, items[0][3])
 
-    # -- PS path (delegates to parallel/ps_synchronizer.py) -----------------
-    def _ps_key(self, sh: ShardPlan) -> str:
-        return sh.name + f"/{sh.slice.start if sh.slice else 0}"
-
-    def _issue_ps_round(self, plan: VarPlan, sh: ShardPlan):
-        from autodist_amd.parallel.ps_synchronizer import PSSynchronizer
-        PSSynchronizer.issue_round(self, plan, sh)
-
-    def _consume_ps_rounds(self, plan: VarPlan, sh: ShardPlan):
-        from autodist_amd.parallel.ps_synchronizer import PSSynchronizer
-        PSSynchronizer.consume_due_rounds(self, plan, sh)
-
     def drain(self):
         """Consume ALL outstanding PS rounds (end of training / checkpoint)."""
-        for plan in self.var_plans:
-            if plan.sparse:
-                continue
-            for sh in plan.shards:
-                if sh.kind == "ps":
-                    key = self._ps_key(sh)
-                    rounds = self._ps_outstanding.get(key, [])
-                    had_rounds = bool(rounds)
-                    while rounds:
-                        r = rounds.pop(0)
-                        if r.event is not None:
-                            torch.cuda.current_stream().wait_event(r.event)
-                        if r.handle is not None:
-                            for h in (r.handle if isinstance(r.handle, tuple)
-                                      else (r.handle,)):
-                                if h is not None:
-                                    h.wait()
-                    if had_rounds:
-                        view = sh.slice.view(plan.param.data) if sh.slice \
-                            else plan.param.data
-                        view.copy_(sh.stage)
+        for grp in self.ps_groups:
+            grp.drain(self)
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 
@@ -941,6 +920,8 @@ class DistributedEngine:
             "n_buckets": len(self.buckets),
             "allreduce_bytes_per_step": ar_bytes,
             "ps_shards": n_ps_shards,
+            "ps_owner_groups": len(self.ps_groups),
+            "ps_collectives_per_step": 2 * len(self.ps_groups),
             "ps_bytes_per_step": ps_bytes,
             "partitioned_ar_shards": n_reducers,
             "partitioned_ar_bytes_per_step": shard_reduce_bytes,

@@ -1,18 +1,29 @@
-"""PS synchronizer — shard-owner parameter-server rounds over xGMI.
+"""PS synchronizer — batched shard-owner parameter-server rounds over xGMI.
 
 Reference behavior: autodist/kernel/synchronization/ps_synchronizer.py (the
 reference's biggest kernel, 761 LoC): local dense aggregation (AddN+RealDiv,
 460-474), cross-worker ConditionalAccumulator with count-gated take_grad
 (556-633), FIFOQueue token barriers for sync (335-385), staleness-bounded
-queues (388-458), ProxyVariable caching (537-554).
+queues (388-458), ProxyVariable caching (537-554). The reference overlaps
+PS traffic with compute through TF dataflow ordering (250-332).
 
 MI355X-native re-derivation (single xGMI node, one rank per GPU):
 
-  * The owner rank holds the MASTER copy of its shard + shard-local
-    optimizer state (replaces PS-device variables + accumulators).
-  * One round = reduce(grad -> owner) + owner fused-apply + broadcast(master)
-    — all enqueued on the engine's comm HIP stream, so rounds pipeline with
-    compute (replaces accumulator/apply/read dataflow).
+  * Shards are COALESCED per (owner, dtype, sync, staleness) into
+    PSOwnerGroups with flat grad/stage/master buffers, so one step costs
+    ONE reduce + ONE broadcast per owner group — O(world) collectives
+    instead of O(variables) (the round-1 design issued a reduce+broadcast
+    pair per shard per step; for BERT-scale PSLoadBalancing that was
+    hundreds of small latency-bound collectives).
+  * Group reduces are ISSUED FROM post-accumulate-grad hooks as soon as the
+    group's last member gradient lands, so PS traffic overlaps backward
+    exactly like the AR buckets (the reference's dataflow overlap).
+  * The owner rank holds the MASTER copy (a view into the group's flat
+    master buffer) + shard-local optimizer state (replaces PS-device
+    variables + accumulators).
+  * One round = reduce(flat_grad -> owner) + owner grouped-apply +
+    broadcast(flat) — all enqueued on the engine's comm HIP stream, so
+    rounds pipeline with compute.
   * SYNC barrier semantics: with staleness=0 the round's broadcast is
     consumed before the next step's forward (the token-queue barrier,
     335-385). With staleness=k, up to k rounds stay in flight — workers run
@@ -20,10 +31,11 @@ MI355X-native re-derivation (single xGMI node, one rank per GPU):
     sync=False rounds are consumed opportunistically up to a queue cap
     (async PS, 261-262).
   * ProxyVariable (537-554): every rank's live parameter IS the local proxy;
-    the post-update broadcast into the staging buffer + stream-ordered copy
-    is the proxy refresh. local_replication=False keeps the same traffic but
-    the refresh is applied lazily at consume time (direct read semantics).
+    the post-update broadcast into the flat staging buffer + stream-ordered
+    copy is the proxy refresh.
 """
+from typing import List, Optional
+
 import torch
 import torch.distributed as dist
 
@@ -33,12 +45,205 @@ ASYNC_PS_MAX_DEPTH = 4  # queue cap for sync=False (unbounded staleness)
 
 
 class PSRound:
-    """One in-flight reduce->apply->broadcast round."""
+    """One in-flight reduce->apply->broadcast round of a whole group."""
 
-    def __init__(self, shard, event=None, handle=None):
-        self.shard = shard
+    def __init__(self, event=None, handle=None):
         self.event = event     # GPU: comm-stream event after broadcast
-        self.handle = handle   # CPU: async work handle
+        self.handle = handle   # CPU: async work handles
+
+
+class PSOwnerGroup:
+    """All PS shards owned by one rank (same dtype/sync/staleness), fused
+    into flat buffers (the PS analog of the AR Bucket)."""
+
+    def __init__(self, owner_rank: int, dtype: torch.dtype,
+                 device: torch.device, sync: bool, staleness: int):
+        self.owner_rank = owner_rank
+        self.dtype = dtype
+        self.device = device
+        self.sync = sync
+        self.staleness = staleness
+        self.members: List[tuple] = []      # (plan, shard, offset, numel)
+        self.numel = 0
+        self.flat_grad: Optional[torch.Tensor] = None
+        self.flat_stage: Optional[torch.Tensor] = None
+        self.flat_master: Optional[torch.Tensor] = None  # owner only
+        self.rounds: List[PSRound] = []
+        self._ready = 0
+        self._issued = False
+        self._grads_in = set()
+
+    def add(self, plan, sh):
+        view = sh.slice.view(plan.param.data) if sh.slice else plan.param.data
+        n = view.numel()
+        self.members.append((plan, sh, self.numel, n))
+        self.numel += n
+
+    def allocate(self, engine):
+        self.flat_grad = torch.zeros(self.numel, dtype=self.dtype,
+                                     device=self.device)
+        self.flat_stage = torch.empty(self.numel, dtype=self.dtype,
+                                      device=self.device)
+        is_owner = engine.rank == self.owner_rank
+        if is_owner:
+            self.flat_master = torch.empty(self.numel, dtype=self.dtype,
+                                           device=self.device)
+        for plan, sh, off, n in self.members:
+            view = sh.slice.view(plan.param.data) if sh.slice \
+                else plan.param.data
+            self.flat_stage[off:off + n].view(view.shape).copy_(view)
+            sh.stage = self.flat_stage[off:off + n].view(view.shape)
+            if is_owner:
+                self.flat_master[off:off + n].view(view.shape).copy_(view)
+                sh.master = self.flat_master[off:off + n].view(view.shape)
+                sh.state = apply_mod.make_state(plan.cls_name, sh.master,
+                                                plan.hyper)
+            sh.owner_group = self  # runtime back-ref for hooks/checkpoint
+
+    # -- per-step protocol -------------------------------------------------
+    def reset(self):
+        self._ready = 0
+        self._issued = False
+        self._grads_in.clear()
+
+    def copy_grad_in(self, plan, sh, off, n):
+        """Stage one shard's gradient into the flat buffer (idempotent per
+        step)."""
+        key = id(sh)
+        if key in self._grads_in:
+            return
+        if plan.param.grad is None:
+            return
+        self._grads_in.add(key)
+        gview = sh.slice.view(plan.param.grad) if sh.slice \
+            else plan.param.grad
+        self.flat_grad[off:off + n].view(gview.shape).copy_(gview)
+        self._ready += 1
+
+    def mark_param_ready(self, engine, plan):
+        """Hook path: stage this plan's shards; issue the group reduce the
+        moment the last member lands (overlaps the rest of backward)."""
+        for p, sh, off, n in self.members:
+            if p is plan:
+                self.copy_grad_in(p, sh, off, n)
+        if self._ready >= len(self.members) and not self._issued:
+            self.issue(engine)
+
+    def _flush_grads(self, engine):
+        for plan, sh, off, n in self.members:
+            self.copy_grad_in(plan, sh, off, n)
+
+    def issue(self, engine):
+        """Scale + reduce the flat gradient toward the owner."""
+        if self._issued:
+            return
+        self._issued = True
+        self._flush_grads(engine)
+        if not engine.collectives_active:
+            return
+        if engine.device.type == "cuda":
+            ev = torch.cuda.Event()
+            ev.record()
+            engine.comm_stream.wait_event(ev)
+            with torch.cuda.stream(engine.comm_stream):
+                self.flat_grad.mul_(engine.grad_scale())
+                self._reduce_handle = dist.reduce(
+                    self.flat_grad, dst=self.owner_rank,
+                    group=engine.process_group, async_op=True)
+        else:
+            self.flat_grad.mul_(engine.grad_scale())
+            self._reduce_handle = dist.reduce(
+                self.flat_grad, dst=self.owner_rank,
+                group=engine.process_group, async_op=True)
+
+    def _grouped_apply(self, engine):
+        """Owner: one multi-tensor apply over every member shard (grouped by
+        optimizer class + hyper — the fused `ResourceApply*` layer)."""
+        groups = {}
+        for plan, sh, off, n in self.members:
+            if plan.param.grad is None and id(sh) not in self._grads_in:
+                continue
+            g = self.flat_grad[off:off + n].view(sh.master.shape)
+            key = (plan.cls_name, plan.group_index, id(plan.hyper))
+            groups.setdefault(key, ([], [], [], plan.hyper))
+            groups[key][0].append(sh.master)
+            groups[key][1].append(g)
+            groups[key][2].append(sh.state)
+        for (cls_name, _, _), (ps, gs, sts, hyper) in groups.items():
+            apply_mod.apply_dense(cls_name, ps, gs, sts, hyper)
+
+    def apply_and_broadcast(self, engine):
+        """Complete the round: owner applies + broadcasts the fresh values;
+        every rank's flat_stage ends up holding them."""
+        if not self._issued:
+            self.issue(engine)
+        is_owner = engine.rank == self.owner_rank
+        if not engine.collectives_active:
+            # local-only: apply directly, stage mirrors master
+            self._grouped_apply(engine)
+            self.flat_stage.copy_(self.flat_master)
+            self.rounds.append(PSRound())
+            return
+        if engine.device.type == "cuda":
+            with torch.cuda.stream(engine.comm_stream):
+                if is_owner:
+                    self._reduce_handle.wait()
+                    self._grouped_apply(engine)
+                    self.flat_stage.copy_(self.flat_master)
+                dist.broadcast(self.flat_stage, src=self.owner_rank,
+                               group=engine.process_group)
+            done = torch.cuda.Event()
+            done.record(engine.comm_stream)
+            self.rounds.append(PSRound(event=done))
+        else:
+            # CPU/gloo: async handles so non-owner workers RUN AHEAD within
+            # the staleness bound (the c9-verified behavior) — only the
+            # owner blocks, on the reduced gradient it must apply.
+            handles = []
+            if is_owner:
+                self._reduce_handle.wait()
+                self._grouped_apply(engine)
+                self.flat_stage.copy_(self.flat_master)
+            h_bc = dist.broadcast(self.flat_stage, src=self.owner_rank,
+                                  group=engine.process_group, async_op=True)
+            handles.append(h_bc)
+            self.rounds.append(PSRound(handle=tuple(handles)))
+
+    def _wait_round(self, r: PSRound):
+        if r.event is not None:
+            torch.cuda.current_stream().wait_event(r.event)
+        if r.handle is not None:
+            for h in r.handle:
+                if h is not None:
+                    h.wait()
+
+    def _install(self):
+        """Copy the (complete) staged values into the live parameters — the
+        proxy refresh."""
+        for plan, sh, off, n in self.members:
+            view = sh.slice.view(plan.param.data) if sh.slice \
+                else plan.param.data
+            view.copy_(sh.stage)
+
+    def consume_due(self, engine):
+        """Pop rounds past the staleness bound and install the newest
+        (reference staleness queues ps_synchronizer.py:388-458; sync token
+        barrier 335-385)."""
+        depth = self.staleness if self.sync else ASYNC_PS_MAX_DEPTH
+        installed = False
+        while len(self.rounds) > depth:
+            r = self.rounds.pop(0)
+            self._wait_round(r)
+            installed = True
+        if installed:
+            self._install()
+
+    def drain(self, engine):
+        had = bool(self.rounds)
+        while self.rounds:
+            self._wait_round(self.rounds.pop(0))
+        if had:
+            self._install()
 
 
 class PSSynchronizer:
@@ -46,80 +251,24 @@ class PSSynchronizer:
     between_graph_apply, ps_synchronizer.py:250-332)."""
 
     @staticmethod
-    def issue_round(engine, plan, sh):
-        grad = plan.param.grad
-        if grad is None:
-            return
-        gview = sh.slice.view(grad) if sh.slice else grad
-        gbuf = gview if gview.is_contiguous() else gview.contiguous()
+    def build_groups(engine, ps_items) -> List[PSOwnerGroup]:
+        """Coalesce (plan, shard) pairs into per-owner flat groups.
 
-        def round_body():
-            gbuf.mul_(engine.grad_scale())
-            dist.reduce(gbuf, dst=sh.owner_rank, group=engine.process_group)
-            if engine.rank == sh.owner_rank:
-                apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
-                                      [sh.state], plan.hyper)
-                src_buf = sh.master
-            else:
-                src_buf = sh.stage
-            dist.broadcast(src_buf, src=sh.owner_rank,
-                           group=engine.process_group)
-            if engine.rank == sh.owner_rank:
-                sh.stage.copy_(sh.master)
-
-        key = engine._ps_key(sh)
-        if not engine.collectives_active:
-            apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
-                                  [sh.state], plan.hyper)
-            sh.stage.copy_(sh.master)
-            engine._ps_outstanding[key].append(PSRound(sh))
-            return
-        if engine.device.type == "cuda":
-            ev = torch.cuda.Event()
-            ev.record()
-            engine.comm_stream.wait_event(ev)
-            with torch.cuda.stream(engine.comm_stream):
-                round_body()
-            done = torch.cuda.Event()
-            done.record(engine.comm_stream)
-            engine._ps_outstanding[key].append(PSRound(sh, event=done))
-        else:
-            # CPU/gloo: async handles so non-owner workers RUN AHEAD within
-            # the staleness bound (the c9-verified behavior) — only the owner
-            # blocks, on the reduced gradient it must apply.
-            gbuf.mul_(engine.grad_scale())
-            h_red = dist.reduce(gbuf, dst=sh.owner_rank,
-                                group=engine.process_group, async_op=True)
-            if engine.rank == sh.owner_rank:
-                h_red.wait()
-                apply_mod.apply_dense(plan.cls_name, [sh.master], [gbuf],
-                                      [sh.state], plan.hyper)
-                sh.stage.copy_(sh.master)
-                src_buf = sh.master
-            else:
-                src_buf = sh.stage
-            h_bc = dist.broadcast(src_buf, src=sh.owner_rank,
-                                  group=engine.process_group, async_op=True)
-            engine._ps_outstanding[key].append(
-                PSRound(sh, handle=(h_red, h_bc)))
-
-    @staticmethod
-    def consume_due_rounds(engine, plan, sh):
-        """Pop rounds past the staleness bound; install results into the live
-        parameter (reference staleness queues, ps_synchronizer.py:388-458;
-        sync token barrier 335-385)."""
-        key = engine._ps_key(sh)
-        rounds = engine._ps_outstanding[key]
-        depth = sh.staleness if sh.sync else ASYNC_PS_MAX_DEPTH
-        while len(rounds) > depth:
-            r = rounds.pop(0)
-            if r.event is not None:
-                torch.cuda.current_stream().wait_event(r.event)
-            if r.handle is not None:
-                for h in (r.handle if isinstance(r.handle, tuple)
-                          else (r.handle,)):
-                    if h is not None:
-                        h.wait()
-            view = sh.slice.view(plan.param.data) if sh.slice \
-                else plan.param.data
-            view.copy_(sh.stage)
+        Deterministic membership/order (sorted by shard name + slice start)
+        so every rank builds identical buffers and enqueues identical
+        collectives."""
+        groups = {}
+        items = sorted(
+            ps_items,
+            key=lambda t: (t[1].name, t[1].slice.start if t[1].slice else 0))
+        for plan, sh in items:
+            key = (sh.owner_rank, plan.param.dtype, sh.sync, sh.staleness)
+            if key not in groups:
+                groups[key] = PSOwnerGroup(sh.owner_rank, plan.param.dtype,
+                                           engine.device, sh.sync,
+                                           sh.staleness)
+            groups[key].add(plan, sh)
+        out = [groups[k] for k in sorted(groups, key=repr)]
+        for g in out:
+            g.allocate(engine)
+        return out

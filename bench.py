@@ -270,11 +270,12 @@ def main():
 
     # hipGraph capture: the per-step kernel chain is launch-bound in eager
     # mode; replaying a captured graph removes host launch + python overhead
-    # entirely. world>1: RCCL collectives are capturable (torch NCCL graph
-    # support); non-graph-safe models (sparse/PS host-side logic, cuDNN RNN)
-    # stay eager unless forced.
+    # entirely. Capturing RCCL collectives SEGFAULTS in this ROCm 7.2 stack
+    # (hipGraph capture_end crash, measured 2026-09-14), so auto-capture is
+    # gated to runs with no active collectives; --hipgraph on forces a try.
     use_graph = use_cuda and (args.hipgraph == "on" or
-                              (args.hipgraph == "auto" and wl["graph_safe"]))
+                              (args.hipgraph == "auto" and wl["graph_safe"]
+                               and not engine.collectives_active))
     run_step = step
     for _ in range(args.warmup):
         step()

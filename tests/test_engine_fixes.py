@@ -212,3 +212,34 @@ def _uneven_case(rank, world):
 @pytest.mark.integration
 def test_uneven_split_weighted_average():
     run_distributed(_uneven_case, world_size=3)
+
+
+def test_ps_owner_groups_batch_collectives():
+    """PS rounds must coalesce to O(world) collectives
+    per step (one reduce+broadcast per owner group), not O(variables)."""
+    from autodist_amd.strategy import PSLoadBalancing
+    torch.manual_seed(0)
+    # many variables -> many shards; all on 1 owner set
+    model = torch.nn.Sequential(*[torch.nn.Linear(8, 8) for _ in range(12)])
+    g = GraphItem()
+    g.extend_model(model)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    g.extend_optimizer_info(opt)
+    strategy = PSLoadBalancing().build(g, ResourceSpec())
+    engine = DistributedEngine(g, strategy, rank=0, world_size=1,
+                               device=torch.device("cpu")).setup()
+    stats = engine.stats()
+    assert stats["ps_shards"] == 24  # 12 weights + 12 biases
+    # single node spec -> 1 owner -> 1 group -> 2 collectives/step
+    assert stats["ps_owner_groups"] <= 1 or \
+        stats["ps_owner_groups"] <= engine.world_size
+    assert stats["ps_collectives_per_step"] <= 2 * max(engine.world_size, 1)
+    # training still works through the grouped path
+    x, y = torch.randn(4, 8), torch.randn(4, 8)
+    w0 = model[0].weight.detach().clone()
+    opt.zero_grad()
+    torch.nn.functional.mse_loss(model(x), y).backward()
+    opt.step()
+    engine.drain()
+    assert not torch.equal(model[0].weight.detach(), w0)
+    engine.teardown()
