@@ -49,7 +49,8 @@ def test_attn_fwd_outlier_rows():
 
 
 def test_fused_sdpa_dispatch():
-    """fused_sdpa uses the kernel under no_grad and falls back under grad."""
+    """fused_sdpa uses the kernel for inference AND training (the backward
+    was GPU-validated in round 2); masked/dropout cases fall back to SDPA."""
     from autodist_amd.ops.fused_attention import can_use_fused, fused_sdpa
     q = torch.randn(1, 2, 64, 64, device="cuda", dtype=torch.bfloat16)
     with torch.no_grad():
@@ -59,7 +60,12 @@ def test_fused_sdpa_dispatch():
             q.float(), q.float(), q.float())
         assert (o.float() - ref).abs().max().item() < 3e-2
     qg = q.clone().requires_grad_(True)
-    assert not can_use_fused(qg, None, 0.0)
+    assert can_use_fused(qg, None, 0.0)          # training path now fused
+    assert not can_use_fused(qg, None, 0.1)      # dropout -> SDPA fallback
+    mask = torch.zeros(1, 1, 64, 64, device="cuda", dtype=torch.bfloat16)
+    assert not can_use_fused(qg, mask, 0.0)      # mask -> SDPA fallback
+    out = fused_sdpa(qg, qg, qg)
+    assert out.grad_fn is not None
 
 
 def test_bert_eval_uses_fused_path():
