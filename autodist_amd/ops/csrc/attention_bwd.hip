@@ -1,8 +1,9 @@
-// MFMA attention BACKWARD (draft — compile-checked; GPU validation is the
-// first round-2 task; the GPU test is gated behind AUTODIST_EXPERIMENTAL).
+// MFMA attention BACKWARD (GPU-validated vs fp32 autograd, round 2).
 //
 // Flash-attention-2-style backward for the forward in attention.hip
-// (bf16, D=64, S%32==0, no mask/dropout):
+// (bf16, D=64, S%32==0; optional additive key mask [B,1,1,S] and
+// hash-counter dropout regenerated from (seed, bh, q, k) — identical to
+// the forward's mask, no S x S state):
 //
 //   delta_i = rowsum(dO_i * O_i)
 //   dV = P^T dO          dP = dO V^T
@@ -25,6 +26,18 @@ typedef short bwd_bf16x8 __attribute__((ext_vector_type(8)));
 typedef float bwd_f32x4 __attribute__((ext_vector_type(4)));
 
 #define ATTN_BD 64
+
+__device__ __forceinline__ unsigned int bwd_drop_hash(unsigned int seed,
+                                                      unsigned int bh,
+                                                      unsigned int q,
+                                                      unsigned int k) {
+  unsigned int x = seed ^ (bh * 0x9E3779B9u) ^ (q * 0x85EBCA6Bu)
+                   ^ (k * 0xC2B2AE35u);
+  x ^= x >> 16; x *= 0x7FEB352Du;
+  x ^= x >> 15; x *= 0x846CA68Bu;
+  x ^= x >> 16;
+  return x;
+}
 
 __device__ __forceinline__ float bwd_red_max(float v) {
 #pragma unroll
@@ -56,8 +69,10 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                                   __hip_bfloat16* __restrict__ dQ,
                                   float* __restrict__ Mbuf,
                                   float* __restrict__ Lbuf,
-                                  float* __restrict__ Dbuf, long S,
-                                  float scale) {
+                                  float* __restrict__ Dbuf,
+                                  const float* __restrict__ mask, long S,
+                                  long H, float scale, float p_drop,
+                                  unsigned int seed) {
   __shared__ float PS[16][32 + 1];
   int l = threadIdx.x;
   long bh = blockIdx.y;
@@ -69,6 +84,11 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
   const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * ATTN_BD;
   short* dq_p = reinterpret_cast<short*>(dQ) + bh * S * ATTN_BD;
   int am = l & 15, kg = l >> 4;
+  const float* m_p = mask ? mask + (bh / H) * S : nullptr;
+  const unsigned int thresh =
+      (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
+  const bool do_drop = p_drop > 0.0f;
+  const float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
 
   bwd_bf16x8 qf[2], dof[2];
 #pragma unroll
@@ -92,9 +112,12 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                                                            s_acc[h], 0, 0, 0);
       }
     }
+    float mv0 = m_p ? m_p[kt + am] : 0.f;
+    float mv1 = m_p ? m_p[kt + 16 + am] : 0.f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float s0 = s_acc[0][r] * scale, s1 = s_acc[1][r] * scale;
+      float s0 = s_acc[0][r] * scale + mv0;
+      float s1 = s_acc[1][r] * scale + mv1;
       float tmax = bwd_red_max(fmaxf(s0, s1));
       float m_new = fmaxf(m_run[r], tmax);
       float alpha = __expf(m_run[r] - m_new);
@@ -146,13 +169,21 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                                                             dp_acc[h], 0, 0, 0);
       }
     }
-    // dS = P * (dP - delta) * scale   (fold dQ's trailing *scale here)
+    // dS = P * (keep*dP/(1-p) - delta) * scale (fold dQ's trailing scale)
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
+      float mv = m_p ? m_p[kt + h * 16 + am] : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = __expf(s_acc[h][r] * scale - m_run[r]) / l_run[r];
-        float ds = p * (dp_acc[h][r] - delta[r]) * scale;
+        float p = __expf(s_acc[h][r] * scale + mv - m_run[r]) / l_run[r];
+        float dp = dp_acc[h][r];
+        if (do_drop) {
+          unsigned int keep = bwd_drop_hash(
+              seed, (unsigned int)bh, (unsigned int)(q0 + kg * 4 + r),
+              (unsigned int)(kt + h * 16 + am)) >= thresh;
+          dp = keep ? dp * rkeep : 0.f;
+        }
+        float ds = p * (dp - delta[r]) * scale;
         PS[kg * 4 + r][h * 16 + am] = ds;
       }
     }
@@ -197,8 +228,10 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
                                    __hip_bfloat16* __restrict__ dV,
                                    const float* __restrict__ Mbuf,
                                    const float* __restrict__ Lbuf,
-                                   const float* __restrict__ Dbuf, long S,
-                                   float scale) {
+                                   const float* __restrict__ Dbuf,
+                                   const float* __restrict__ mask, long S,
+                                   long H, float scale, float p_drop,
+                                   unsigned int seed) {
   __shared__ float PS[16][32 + 1];   // P' or dS' tile [key][q-chunk]
   __shared__ float PS2[16][32 + 1];
   int l = threadIdx.x;
@@ -211,6 +244,11 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
   short* dk_p = reinterpret_cast<short*>(dK) + bh * S * ATTN_BD;
   short* dv_p = reinterpret_cast<short*>(dV) + bh * S * ATTN_BD;
   int am = l & 15, kg = l >> 4;
+  const float* m_p = mask ? mask + (bh / H) * S : nullptr;
+  const unsigned int thresh =
+      (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
+  const bool do_drop = p_drop > 0.0f;
+  const float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
 
   bwd_bf16x8 kf[2], vf[2];
 #pragma unroll
@@ -249,9 +287,18 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
       float d_q = Dbuf[bh * S + qrow];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = __expf(s_acc[h][r] * scale - m_q) / l_q;
-        PS[kg * 4 + r][h * 16 + am] = p;                        // P'
-        PS2[kg * 4 + r][h * 16 + am] = p * (dp_acc[h][r] - d_q) * scale;
+        float mv = m_p ? m_p[k0 + kg * 4 + r] : 0.f;  // key-row mask
+        float p = __expf(s_acc[h][r] * scale + mv - m_q) / l_q;
+        float pd = p, dp = dp_acc[h][r];
+        if (do_drop) {
+          unsigned int keep = bwd_drop_hash(
+              seed, (unsigned int)bh, (unsigned int)qrow,
+              (unsigned int)(k0 + kg * 4 + r)) >= thresh;
+          pd = keep ? p * rkeep : 0.f;   // dropped P' for dV
+          dp = keep ? dp * rkeep : 0.f;  // dropped dP' for dK
+        }
+        PS[kg * 4 + r][h * 16 + am] = pd;                       // P'
+        PS2[kg * 4 + r][h * 16 + am] = p * (dp - d_q) * scale;
       }
     }
     __syncthreads();

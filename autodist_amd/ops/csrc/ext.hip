@@ -417,8 +417,20 @@ void psgd_decompress_ef(torch::Tensor flat, torch::Tensor err,
                      Qp.data_ptr<float>(), numel, s, (float)scale);
 }
 
+static const float* attn_mask_ptr(const c10::optional<torch::Tensor>& mask,
+                                  long B, long S) {
+  if (!mask.has_value() || !mask->defined()) return nullptr;
+  TORCH_CHECK(mask->scalar_type() == torch::kFloat32 && mask->is_contiguous(),
+              "attention mask must be contiguous fp32");
+  TORCH_CHECK(mask->numel() == B * S,
+              "attention mask must be an additive [B,1,1,S] key mask");
+  return mask->data_ptr<float>();
+}
+
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                       double scale) {
+                       double scale,
+                       c10::optional<torch::Tensor> mask = c10::nullopt,
+                       double p_drop = 0.0, int64_t seed = 0) {
   TORCH_CHECK(q.dim() == 4 && q.size(3) == ATTN_D,
               "attn_fwd expects [B,H,S,64]");
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
@@ -427,19 +439,33 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(S % 32 == 0 && S >= 32, "S must be a multiple of 32");
   TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes());
   auto o = torch::empty_like(q);
-  dim3 grid(S / 16, B * H);
-  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(64), 0, cur_stream(),
+  dim3 grid((S + 63) / 64, B * H);
+  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, cur_stream(),
                      reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
                      reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
                      reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
-                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()), S,
-                     (float)scale);
+                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
+                     attn_mask_ptr(mask, B, S), S, H, (float)scale,
+                     (float)p_drop, (unsigned int)(uint64_t)seed);
   return o;
+}
+
+torch::Tensor attn_dropmask(long B, long H, long S, double p_drop,
+                            int64_t seed, torch::Device dev) {
+  auto out = torch::empty({B, H, S, S},
+                          torch::dtype(torch::kUInt8).device(dev));
+  hipLaunchKernelGGL(attn_dropmask_kernel, dim3(S, B * H), dim3(256), 0,
+                     cur_stream(), out.data_ptr<unsigned char>(), S,
+                     (float)p_drop, (unsigned int)(uint64_t)seed);
+  return out;
 }
 
 std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, torch::Tensor o,
-                                    torch::Tensor dout, double scale) {
+                                    torch::Tensor dout, double scale,
+                                    c10::optional<torch::Tensor> mask
+                                        = c10::nullopt,
+                                    double p_drop = 0.0, int64_t seed = 0) {
   TORCH_CHECK(q.dim() == 4 && q.size(3) == ATTN_BD);
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   long B = q.size(0), H = q.size(1), S = q.size(2);
@@ -451,17 +477,21 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto Mbuf = torch::empty({B * H * S}, fopt);
   auto Lbuf = torch::empty({B * H * S}, fopt);
   auto Dbuf = torch::empty({B * H * S}, fopt);
+  const float* mp = attn_mask_ptr(mask, B, S);
+  unsigned int sd = (unsigned int)(uint64_t)seed;
   dim3 grid(S / 16, B * H);
   auto st = cur_stream();
 #define BF16P(t) reinterpret_cast<__hip_bfloat16*>((t).data_ptr())
   hipLaunchKernelGGL(attn_bwd_q_kernel, grid, dim3(64), 0, st, BF16P(q),
                      BF16P(k), BF16P(v), BF16P(o), BF16P(dout), BF16P(dq),
                      Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
-                     Dbuf.data_ptr<float>(), S, (float)scale);
+                     Dbuf.data_ptr<float>(), mp, S, H, (float)scale,
+                     (float)p_drop, sd);
   hipLaunchKernelGGL(attn_bwd_kv_kernel, grid, dim3(64), 0, st, BF16P(q),
                      BF16P(k), BF16P(v), BF16P(dout), BF16P(dk), BF16P(dv),
                      Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
-                     Dbuf.data_ptr<float>(), S, (float)scale);
+                     Dbuf.data_ptr<float>(), mp, S, H, (float)scale,
+                     (float)p_drop, sd);
 #undef BF16P
   return {dq, dk, dv};
 }
@@ -512,9 +542,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "padded M = flat + err (zero tail)");
   m.def("mfma_probe", &mfma_probe,
         "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
+  m.def("attn_dropmask", &attn_dropmask,
+        "materialize the hash dropout keep-mask (testing)");
   m.def("attn_fwd", &attn_fwd,
-        "fused MFMA attention forward (bf16, D=64, no mask) — serving path");
+        "fused MFMA attention forward (bf16, D=64; optional additive key "
+        "mask + hash dropout) — 4-wave LDS-tiled",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
+        py::arg("mask") = py::none(), py::arg("p_drop") = 0.0,
+        py::arg("seed") = 0);
   m.def("attn_bwd", &attn_bwd,
-        "fused MFMA attention backward (DRAFT; validate before use) -> "
-        "dq, dk, dv");
+        "fused MFMA attention backward (GPU-validated) -> dq, dk, dv",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"),
+        py::arg("dout"), py::arg("scale"), py::arg("mask") = py::none(),
+        py::arg("p_drop") = 0.0, py::arg("seed") = 0);
 }

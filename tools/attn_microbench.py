@@ -68,7 +68,16 @@ def main():
             torch.autograd.grad(o, (qg, kg, vg), dout)
 
         def fused_fb():
-            o = FusedAttentionFn.apply(qg, kg, vg, scale)
+            o = FusedAttentionFn.apply(qg, kg, vg, scale, None, 0.0, 0)
+            torch.autograd.grad(o, (qg, kg, vg), dout)
+
+        def sdpa_fb_drop():
+            o = torch.nn.functional.scaled_dot_product_attention(
+                qg, kg, vg, scale=scale, dropout_p=0.1)
+            torch.autograd.grad(o, (qg, kg, vg), dout)
+
+        def fused_fb_drop():
+            o = FusedAttentionFn.apply(qg, kg, vg, scale, None, 0.1, 777)
             torch.autograd.grad(o, (qg, kg, vg), dout)
 
         row = {
@@ -77,9 +86,13 @@ def main():
             "fused_fwd_ms": round(bench(fused_fwd, args.iters), 4),
             "sdpa_fb_ms": round(bench(sdpa_fb, args.iters), 4),
             "fused_fb_ms": round(bench(fused_fb, args.iters), 4),
+            "sdpa_fb_drop_ms": round(bench(sdpa_fb_drop, args.iters), 4),
+            "fused_fb_drop_ms": round(bench(fused_fb_drop, args.iters), 4),
         }
         row["fwd_speedup"] = round(row["sdpa_fwd_ms"] / row["fused_fwd_ms"], 3)
         row["fb_speedup"] = round(row["sdpa_fb_ms"] / row["fused_fb_ms"], 3)
+        row["fb_drop_speedup"] = round(
+            row["sdpa_fb_drop_ms"] / row["fused_fb_drop_ms"], 3)
         results.append(row)
         print(row, flush=True)
     import json
