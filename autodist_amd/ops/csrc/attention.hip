@@ -6,18 +6,21 @@
 // hash of (seed, bh, q, k) so the backward regenerates the identical mask
 // without materializing S x S state (flash-attention-style).
 //
-// v2 structure (4 waves x 16-query rows = 64-query tile per block):
-//   * K/V 32-key tiles are staged in LDS ONCE per block and consumed by all
-//     4 waves -> 4x less L2/HBM K/V traffic than the round-1 one-wave
-//     kernel (which re-read K/V per 16-row tile and was 0.5-0.9x SDPA).
-//   * QK^T and P@V on v_mfma_f32_16x16x32_bf16. A/B fragments use the
-//     contiguous-8 k-map; per the measured probe (tests/test_mfma_probe.py)
-//     any SELF-CONSISTENT A/B k-map is valid, and P is routed through a
-//     per-wave LDS buffer (C-layout write, A-layout read).
-//   * online softmax in fp32: per-row running max m / sum l, row reductions
-//     via 4-step shfl_xor over the 16-lane C-column group. l accumulates
-//     the UNdropped probabilities (torch semantics: dropout after softmax);
-//     the 1/(1-p) rescale is folded into the epilogue.
+// v3 structure (4 waves x 16-query rows = 64-query tile per block):
+//   * 64-KEY K/V tiles staged in DOUBLE-BUFFERED LDS, consumed by all 4
+//     waves: one block barrier per 64 keys (the v2 design paid 3 barriers
+//     per 32 keys and was barrier-bound at S >= 512). Next tile's global
+//     loads are issued into the other buffer before computing the current
+//     one, so HBM latency overlaps the MFMA work.
+//   * V is staged TRANSPOSED so P@V B-fragments are single ds_read_b128s;
+//     P is staged bf16 per wave (C-layout write, A-layout read) with a
+//     wave-local s_waitcnt fence instead of a block barrier.
+//   * QK^T and P@V on v_mfma_f32_16x16x32_bf16 (contiguous-8 k-map; see
+//     tests/test_mfma_probe.py for the measured layout contract).
+//   * online softmax in fp32: per-row running max m / sum l, row
+//     reductions via shfl_xor over the 16-lane C-column group. l
+//     accumulates the UNdropped probabilities (torch semantics: dropout
+//     after softmax); the 1/(1-p) rescale is folded into the epilogue.
 #include "common.h"
 
 typedef short bf16x8_t __attribute__((ext_vector_type(8)));
@@ -27,7 +30,6 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define KPAD 8  // LDS row padding (shorts) to stagger banks
 
 __device__ __forceinline__ float row_reduce_max(float v, int width16) {
-  // max across the 16-lane group (lanes sharing l>>4)
 #pragma unroll
   for (int off = 1; off < 16; off <<= 1) {
     v = fmaxf(v, __shfl_xor(v, off, 64));
@@ -41,6 +43,13 @@ __device__ __forceinline__ float row_reduce_sum(float v) {
     v += __shfl_xor(v, off, 64);
   }
   return v;
+}
+
+// order LDS writes before reads WITHIN one wave (cheaper than a block
+// barrier; wavefront lockstep makes it safe once the counters drain)
+__device__ __forceinline__ void wave_lds_fence() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_wave_barrier();
 }
 
 // counter-based dropout hash: uniform in [0, 2^32); keep iff >= p*2^32.
@@ -59,17 +68,18 @@ __device__ __forceinline__ unsigned int drop_hash(unsigned int seed,
 }
 
 // grid (ceil(S/64), B*H), 256 threads (4 waves). mask may be nullptr.
-__global__ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
-                                const __hip_bfloat16* __restrict__ K,
-                                const __hip_bfloat16* __restrict__ V,
-                                __hip_bfloat16* __restrict__ O,
-                                const float* __restrict__ mask, long S,
-                                long H, float scale, float p_drop,
-                                unsigned int seed) {
-  __shared__ short Ks[32][ATTN_D + KPAD];
-  __shared__ short VsT[ATTN_D][32 + KPAD];  // transposed: B-frag reads are
-                                            // row-contiguous (1 ds_read_b128)
-  __shared__ float P[4][16][32 + 1];
+__global__ void
+__launch_bounds__(256, 2)
+attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
+                const __hip_bfloat16* __restrict__ K,
+                const __hip_bfloat16* __restrict__ V,
+                __hip_bfloat16* __restrict__ O,
+                const float* __restrict__ mask, long S, long H, float scale,
+                float p_drop, unsigned int seed) {
+  __shared__ short Ks[2][64][ATTN_D + KPAD];
+  __shared__ short VsT[2][ATTN_D][64 + KPAD];  // transposed: B-frag reads
+                                               // are one ds_read_b128
+  __shared__ short Pw[4][16][64 + KPAD];       // per-wave bf16 P staging
   int t = threadIdx.x;
   int w = t >> 6;        // wave 0..3
   int l = t & 63;        // lane
@@ -108,61 +118,88 @@ __global__ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt) o_acc[dt] = {0.f, 0.f, 0.f, 0.f};
 
-  // cooperative K/V stage indices: 256 threads x bf16x8 = one 32x64 tile
-  int srow = t >> 3, scol = (t & 7) * 8;
+  // cooperative 64x64 K/V stage: 256 threads x (one row of 64 shorts each
+  // as 2 bf16x8 loads) -> thread t stages row (t>>2) cols (t&3)*16..+15
+  int srow = t >> 2, scol = (t & 3) * 16;
 
-  for (long kt = 0; kt < S; kt += 32) {
-    __syncthreads();  // previous tile's consumers done
-    *reinterpret_cast<bf16x8_t*>(&Ks[srow][scol]) =
-        *reinterpret_cast<const bf16x8_t*>(&k_p[(kt + srow) * ATTN_D + scol]);
-    {
-      bf16x8_t vrow = *reinterpret_cast<const bf16x8_t*>(
-          &v_p[(kt + srow) * ATTN_D + scol]);
+  bf16x8_t kreg[2], vreg[2];
+  auto load_tile = [&](long kt) {  // issue global loads into registers
+    long krow = kt + srow < S ? kt + srow : S - 1;  // clamped; masked later
 #pragma unroll
-      for (int j = 0; j < 8; ++j) VsT[scol + j][srow] = vrow[j];
+    for (int half = 0; half < 2; ++half) {
+      kreg[half] = *reinterpret_cast<const bf16x8_t*>(
+          &k_p[krow * ATTN_D + scol + half * 8]);
+      vreg[half] = *reinterpret_cast<const bf16x8_t*>(
+          &v_p[krow * ATTN_D + scol + half * 8]);
     }
-    __syncthreads();
-
-    // ---- S tile = Q[16] x K[32]^T : two 16x16 C tiles (key halves)
-    f32x4_t s_acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  };
+  auto store_tile = [&](int buf) {  // registers -> LDS (after compute)
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int half = 0; half < 2; ++half) {
+      *reinterpret_cast<bf16x8_t*>(&Ks[buf][srow][scol + half * 8]) =
+          kreg[half];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        VsT[buf][scol + half * 8 + j][srow] = vreg[half][j];
+    }
+  };
+
+  load_tile(0);
+  store_tile(0);
+  int cur = 0;
+  for (long kt = 0; kt < S; kt += 64) {
+    __syncthreads();  // current buffer staged; previous reads done
+    bool has_next = kt + 64 < S;
+    if (has_next) load_tile(kt + 64);  // overlap HBM with the MFMAs below
+
+    // ---- S tile = Q[16] x K[64]^T : four 16x16 C tiles (key quarters)
+    f32x4_t s_acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                        {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
         bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
-            &Ks[h * 16 + am][c * 32 + kg * 8]);
+            &Ks[cur][h * 16 + am][c * 32 + kg * 8]);
         s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kf,
                                                            s_acc[h], 0, 0, 0);
       }
     }
-    // additive key mask (same value for every q row / r)
-    float mv0 = 0.f, mv1 = 0.f;
-    if (m_p) {
-      mv0 = m_p[kt + am];
-      mv1 = m_p[kt + 16 + am];
+    // scale + additive key mask + S-bound masking, then online softmax
+    float sv[4][4];  // [h][r]
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
+      long key = kt + h * 16 + am;
+      float mv = (m_p && key < S) ? m_p[key] : 0.f;
+      float oob = key < S ? 0.f : -1e30f;  // partial last tile
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        sv[h][r] = s_acc[h][r] * scale + mv + oob;
     }
-    // scale + online softmax bookkeeping (row q = kg*4 + r)
-    float p0[4], p1[4], alpha[4];
+    float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float s0 = s_acc[0][r] * scale + mv0;
-      float s1 = s_acc[1][r] * scale + mv1;
-      float tmax = row_reduce_max(fmaxf(s0, s1), 16);
+      float smax = fmaxf(fmaxf(sv[0][r], sv[1][r]),
+                         fmaxf(sv[2][r], sv[3][r]));
+      float tmax = row_reduce_max(smax, 16);
       float m_new = fmaxf(m_run[r], tmax);
       alpha[r] = __expf(m_run[r] - m_new);
-      p0[r] = __expf(s0 - m_new);
-      p1[r] = __expf(s1 - m_new);
-      float rsum = row_reduce_sum(p0[r] + p1[r]);
-      l_run[r] = l_run[r] * alpha[r] + rsum;  // UNdropped sum
+      float psum = 0.f;
+#pragma unroll
+      for (int h = 0; h < 4; ++h) {
+        sv[h][r] = __expf(sv[h][r] - m_new);
+        psum += sv[h][r];
+      }
+      l_run[r] = l_run[r] * alpha[r] + row_reduce_sum(psum);
       m_run[r] = m_new;
       if (do_drop) {
         unsigned int qrow = (unsigned int)(q0 + kg * 4 + r);
-        if (drop_hash(seed, (unsigned int)bh, qrow,
-                      (unsigned int)(kt + am)) < thresh)
-          p0[r] = 0.f;
-        if (drop_hash(seed, (unsigned int)bh, qrow,
-                      (unsigned int)(kt + 16 + am)) < thresh)
-          p1[r] = 0.f;
+#pragma unroll
+        for (int h = 0; h < 4; ++h) {
+          if (drop_hash(seed, (unsigned int)bh, qrow,
+                        (unsigned int)(kt + h * 16 + am)) < thresh)
+            sv[h][r] = 0.f;
+        }
       }
     }
 #pragma unroll
@@ -170,27 +207,36 @@ __global__ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha[r];
     }
-    // ---- stage (dropped) P through per-wave LDS: C-write, A-read
+    // ---- stage (dropped) P bf16 through per-wave LDS: C-write, A-read
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      P[w][kg * 4 + r][am] = p0[r];
-      P[w][kg * 4 + r][16 + am] = p1[r];
-    }
-    __syncthreads();
-    bf16x8_t pf;
+    for (int h = 0; h < 4; ++h) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      __hip_bfloat16 pb = __float2bfloat16(P[w][am][kg * 8 + j]);
-      pf[j] = reinterpret_cast<short&>(pb);
+      for (int r = 0; r < 4; ++r) {
+        __hip_bfloat16 pb = __float2bfloat16(sv[h][r]);
+        Pw[w][kg * 4 + r][h * 16 + am] = reinterpret_cast<short&>(pb);
+      }
     }
-    // ---- O += P @ V : one mfma per 16-col d tile (k = 32 keys)
+    wave_lds_fence();
+    bf16x8_t pf[2];
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      pf[kc] = *reinterpret_cast<const bf16x8_t*>(
+          &Pw[w][am][kc * 32 + kg * 8]);
+    }
+    // ---- O += P @ V : per 16-col d tile, two k-chunks of 32 keys
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
-      bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-          &VsT[dt * 16 + am][kg * 8]);
-      o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf,
-                                                          o_acc[dt], 0, 0, 0);
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc) {
+        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            &VsT[cur][dt * 16 + am][kc * 32 + kg * 8]);
+        o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pf[kc], vf, o_acc[dt], 0, 0, 0);
+      }
     }
+    wave_lds_fence();  // Pw reused next iteration (wave-private)
+    if (has_next) store_tile(1 - cur);
+    cur = 1 - cur;
   }
   // ---- epilogue: normalize (+ dropout keep-rescale) + store
   float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;

@@ -60,10 +60,12 @@ def test_fused_sdpa_dispatch():
             q.float(), q.float(), q.float())
         assert (o.float() - ref).abs().max().item() < 3e-2
     qg = q.clone().requires_grad_(True)
-    assert can_use_fused(qg, None, 0.0)          # training path now fused
-    assert not can_use_fused(qg, None, 0.1)      # dropout -> SDPA fallback
-    mask = torch.zeros(1, 1, 64, 64, device="cuda", dtype=torch.bfloat16)
-    assert not can_use_fused(qg, mask, 0.0)      # mask -> SDPA fallback
+    assert can_use_fused(qg, None, 0.0)       # training path fused
+    assert can_use_fused(qg, None, 0.1)       # hash dropout supported
+    pad = torch.ones(1, 1, 1, 64, device="cuda", dtype=torch.bool)
+    assert can_use_fused(qg, pad, 0.1)        # key-padding mask supported
+    full = torch.zeros(1, 2, 64, 64, device="cuda", dtype=torch.bfloat16)
+    assert not can_use_fused(qg, full, 0.0)   # full S x S mask -> SDPA
     out = fused_sdpa(qg, qg, qg)
     assert out.grad_fn is not None
 
