@@ -29,21 +29,6 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 #define ATTN_D 64
 #define KPAD 8  // LDS row padding (shorts) to stagger banks
 
-__device__ __forceinline__ float row_reduce_max(float v, int width16) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) {
-    v = fmaxf(v, __shfl_xor(v, off, 64));
-  }
-  return v;
-}
-
-__device__ __forceinline__ float row_reduce_sum(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) {
-    v += __shfl_xor(v, off, 64);
-  }
-  return v;
-}
 
 // order LDS writes before reads WITHIN one wave (cheaper than a block
 // barrier; wavefront lockstep makes it safe once the counters drain)
@@ -96,6 +81,7 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   const unsigned int thresh =
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
   const bool do_drop = p_drop > 0.0f;
+  const float scale2 = scale * ATTN_LOG2E;  // exp2-domain logit scale
 
   // Q fragments: q row (q0+am) clamped for partial tiles; d = c*32+kg*8+j
   long qrow_a = q0 + am < S ? q0 + am : S - 1;
@@ -165,32 +151,34 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                                                            s_acc[h], 0, 0, 0);
       }
     }
-    // scale + additive key mask + S-bound masking, then online softmax
+    // scale + additive key mask + S-bound masking, then online softmax.
+    // Everything runs in the exp2 (log2) domain: v_exp_f32 IS exp2, so
+    // exp2f(x) is one instruction where __expf costs an extra multiply.
     float sv[4][4];  // [h][r]
 #pragma unroll
     for (int h = 0; h < 4; ++h) {
-      long key = kt + h * 16 + am;
-      float mv = (m_p && key < S) ? m_p[key] : 0.f;
-      float oob = key < S ? 0.f : -1e30f;  // partial last tile
+      int key = (int)kt + h * 16 + am;
+      float mv = (m_p && key < (int)S) ? m_p[key] * ATTN_LOG2E : 0.f;
+      float oob = key < (int)S ? 0.f : -1e30f;  // partial last tile
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        sv[h][r] = s_acc[h][r] * scale + mv + oob;
+        sv[h][r] = s_acc[h][r] * scale2 + mv + oob;
     }
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float smax = fmaxf(fmaxf(sv[0][r], sv[1][r]),
                          fmaxf(sv[2][r], sv[3][r]));
-      float tmax = row_reduce_max(smax, 16);
+      float tmax = dpp16_max(smax);
       float m_new = fmaxf(m_run[r], tmax);
-      alpha[r] = __expf(m_run[r] - m_new);
+      alpha[r] = exp2f(m_run[r] - m_new);
       float psum = 0.f;
 #pragma unroll
       for (int h = 0; h < 4; ++h) {
-        sv[h][r] = __expf(sv[h][r] - m_new);
+        sv[h][r] = exp2f(sv[h][r] - m_new);
         psum += sv[h][r];
       }
-      l_run[r] = l_run[r] * alpha[r] + row_reduce_sum(psum);
+      l_run[r] = l_run[r] * alpha[r] + dpp16_sum(psum);
       m_run[r] = m_new;
       if (do_drop) {
         unsigned int qrow = (unsigned int)(q0 + kg * 4 + r);

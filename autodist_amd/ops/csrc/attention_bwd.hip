@@ -39,16 +39,9 @@ __device__ __forceinline__ unsigned int bwd_drop_hash(unsigned int seed,
   return x;
 }
 
-__device__ __forceinline__ float bwd_red_max(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
-  return v;
-}
-__device__ __forceinline__ float bwd_red_sum(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, 64);
-  return v;
-}
+// 16-lane reductions via DPP (common.h) — no LDS-pipe traffic
+#define bwd_red_max dpp16_max
+#define bwd_red_sum dpp16_sum
 
 // load an A/B fragment row-block: elem j from src[(row)*64 + c*32 + kg*8+j]
 __device__ __forceinline__ bwd_bf16x8 frag_rowmajor(const short* src,
@@ -89,6 +82,8 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
   const bool do_drop = p_drop > 0.0f;
   const float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const float scale2 = scale * ATTN_LOG2E;  // Mbuf/Lbuf stats are in the
+                                            // exp2 (log2) domain
 
   bwd_bf16x8 qf[2], dof[2];
 #pragma unroll
@@ -112,16 +107,16 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                                                            s_acc[h], 0, 0, 0);
       }
     }
-    float mv0 = m_p ? m_p[kt + am] : 0.f;
-    float mv1 = m_p ? m_p[kt + 16 + am] : 0.f;
+    float mv0 = m_p ? m_p[kt + am] * ATTN_LOG2E : 0.f;
+    float mv1 = m_p ? m_p[kt + 16 + am] * ATTN_LOG2E : 0.f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float s0 = s_acc[0][r] * scale + mv0;
-      float s1 = s_acc[1][r] * scale + mv1;
+      float s0 = s_acc[0][r] * scale2 + mv0;
+      float s1 = s_acc[1][r] * scale2 + mv1;
       float tmax = bwd_red_max(fmaxf(s0, s1));
       float m_new = fmaxf(m_run[r], tmax);
-      float alpha = __expf(m_run[r] - m_new);
-      float rsum = bwd_red_sum(__expf(s0 - m_new) + __expf(s1 - m_new));
+      float alpha = exp2f(m_run[r] - m_new);
+      float rsum = bwd_red_sum(exp2f(s0 - m_new) + exp2f(s1 - m_new));
       l_run[r] = l_run[r] * alpha + rsum;
       m_run[r] = m_new;
     }
@@ -172,10 +167,10 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
     // dS = P * (keep*dP/(1-p) - delta) * scale (fold dQ's trailing scale)
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      float mv = m_p ? m_p[kt + h * 16 + am] : 0.f;
+      float mv = m_p ? m_p[kt + h * 16 + am] * ATTN_LOG2E : 0.f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = __expf(s_acc[h][r] * scale + mv - m_run[r]) / l_run[r];
+        float p = exp2f(s_acc[h][r] * scale2 + mv - m_run[r]) / l_run[r];
         float dp = dp_acc[h][r];
         if (do_drop) {
           unsigned int keep = bwd_drop_hash(
@@ -249,6 +244,8 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
   const bool do_drop = p_drop > 0.0f;
   const float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
+  const float scale2 = scale * ATTN_LOG2E;  // Mbuf/Lbuf stats are in the
+                                            // exp2 (log2) domain
 
   bwd_bf16x8 kf[2], vf[2];
 #pragma unroll
@@ -287,8 +284,8 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
       float d_q = Dbuf[bh * S + qrow];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float mv = m_p ? m_p[k0 + kg * 4 + r] : 0.f;  // key-row mask
-        float p = __expf(s_acc[h][r] * scale + mv - m_q) / l_q;
+        float mv = m_p ? m_p[k0 + kg * 4 + r] * ATTN_LOG2E : 0.f;  // key mask
+        float p = exp2f(s_acc[h][r] * scale2 + mv - m_q) / l_q;
         float pd = p, dp = dp_acc[h][r];
         if (do_drop) {
           unsigned int keep = bwd_drop_hash(

@@ -28,3 +28,44 @@ static inline int grid_for(long n, int per_thread = 4) {
                                hipGetErrorString(e));                    \
     }                                                                    \
   } while (0)
+
+// ---- 16-lane cross-lane reductions via DPP row_ror (VALU-only) ----
+// The 16-lane "row" of DPP matches the MFMA C-column group exactly; four
+// rotate-accumulate steps give every lane the full reduction without any
+// ds_bpermute traffic (shfl_xor lowers to LDS-pipe ops which contended
+// with fragment reads/writes — measured 20:1 VALU:MFMA before this).
+__device__ __forceinline__ float dpp16_sum(float v) {
+  int x;
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x121, 0xF, 0xF, true);  // row_ror:1
+  v += __builtin_bit_cast(float, x);
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x122, 0xF, 0xF, true);  // row_ror:2
+  v += __builtin_bit_cast(float, x);
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x124, 0xF, 0xF, true);  // row_ror:4
+  v += __builtin_bit_cast(float, x);
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x128, 0xF, 0xF, true);  // row_ror:8
+  v += __builtin_bit_cast(float, x);
+  return v;
+}
+
+__device__ __forceinline__ float dpp16_max(float v) {
+  int x;
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x121, 0xF, 0xF, true);
+  v = fmaxf(v, __builtin_bit_cast(float, x));
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x122, 0xF, 0xF, true);
+  v = fmaxf(v, __builtin_bit_cast(float, x));
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x124, 0xF, 0xF, true);
+  v = fmaxf(v, __builtin_bit_cast(float, x));
+  x = __builtin_amdgcn_update_dpp(0, __builtin_bit_cast(int, v),
+                                  0x128, 0xF, 0xF, true);
+  v = fmaxf(v, __builtin_bit_cast(float, x));
+  return v;
+}
+
+#define ATTN_LOG2E 1.4426950408889634f
