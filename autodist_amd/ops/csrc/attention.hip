@@ -52,15 +52,24 @@ __device__ __forceinline__ unsigned int drop_hash(unsigned int seed,
   return x;
 }
 
-// grid (ceil(S/64), B*H), 256 threads (4 waves). mask may be nullptr.
+// 256 threads (4 waves); each wave owns RB x 16 query rows, so a block
+// covers 64*RB queries. K/V tiles are shared by all 4 waves from LDS, so
+// doubling RB HALVES the K/V re-read traffic per query — the kernel is
+// K/V-bandwidth-bound at S >= 512 (measured: SQ busy 4.5%, ~2 TB/s of
+// tile re-reads; SDPA's advantage was exactly its larger q-tile).
+// The grid is launched 1-D as qb * (B*H) + bh: consecutive workgroup ids
+// round-robin across the 8 XCDs, so all q-blocks of one (b,h) — which
+// re-read the SAME K/V — land on the SAME XCD's L2 when B*H % 8 == 0
+// (XCD-aware swizzle; B*H is the fast dimension mod 8).
+template <int RB>
 __global__ void
 __launch_bounds__(256, 2)
 attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                 const __hip_bfloat16* __restrict__ K,
                 const __hip_bfloat16* __restrict__ V,
                 __hip_bfloat16* __restrict__ O,
-                const float* __restrict__ mask, long S, long H, float scale,
-                float p_drop, unsigned int seed) {
+                const float* __restrict__ mask, long S, long H, long NBH,
+                float scale, float p_drop, unsigned int seed) {
   __shared__ short Ks[2][64][ATTN_D + KPAD];
   __shared__ short VsT[2][ATTN_D][64 + KPAD];  // transposed: B-frag reads
                                                // are one ds_read_b128
@@ -68,8 +77,9 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   int t = threadIdx.x;
   int w = t >> 6;        // wave 0..3
   int l = t & 63;        // lane
-  long bh = blockIdx.y;
-  long q0 = (long)blockIdx.x * 64 + w * 16;  // this wave's 16 q rows
+  long bh = (long)blockIdx.x % NBH;            // XCD swizzle (see above)
+  long qb = (long)blockIdx.x / NBH;
+  long q0 = qb * (64 * RB) + w * (16 * RB);    // this wave's RB*16 q rows
   const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * ATTN_D;
   const short* k_p = reinterpret_cast<const short*>(K) + bh * S * ATTN_D;
   const short* v_p = reinterpret_cast<const short*>(V) + bh * S * ATTN_D;
@@ -83,26 +93,32 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   const bool do_drop = p_drop > 0.0f;
   const float scale2 = scale * ATTN_LOG2E;  // exp2-domain logit scale
 
-  // Q fragments: q row (q0+am) clamped for partial tiles; d = c*32+kg*8+j
-  long qrow_a = q0 + am < S ? q0 + am : S - 1;
-  bf16x8_t qf[2];
+  // Q fragments: q row (q0+rb*16+am) clamped for partial tiles
+  bf16x8_t qf[RB][2];
 #pragma unroll
-  for (int c = 0; c < 2; ++c) {
+  for (int rb = 0; rb < RB; ++rb) {
+    int qrow_a = q0 + rb * 16 + am < S ? (int)q0 + rb * 16 + am : (int)S - 1;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      qf[c][j] = q_p[qrow_a * ATTN_D + c * 32 + kg * 8 + j];
+    for (int c = 0; c < 2; ++c) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        qf[rb][c][j] = q_p[qrow_a * ATTN_D + c * 32 + kg * 8 + j];
+      }
     }
   }
 
-  float m_run[4], l_run[4];
-  f32x4_t o_acc[4];  // d-tiles of 16 cols each
+  float m_run[RB][4], l_run[RB][4];
+  f32x4_t o_acc[RB][4];  // d-tiles of 16 cols each
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m_run[r] = -1e30f;
-    l_run[r] = 0.f;
+  for (int rb = 0; rb < RB; ++rb) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_run[rb][r] = -1e30f;
+      l_run[rb][r] = 0.f;
+    }
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) o_acc[rb][dt] = {0.f, 0.f, 0.f, 0.f};
   }
-#pragma unroll
-  for (int dt = 0; dt < 4; ++dt) o_acc[dt] = {0.f, 0.f, 0.f, 0.f};
 
   // cooperative 64x64 K/V stage: 256 threads x (one row of 64 shorts each
   // as 2 bf16x8 loads) -> thread t stages row (t>>2) cols (t&3)*16..+15
@@ -110,7 +126,7 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 
   bf16x8_t kreg[2], vreg[2];
   auto load_tile = [&](long kt) {  // issue global loads into registers
-    long krow = kt + srow < S ? kt + srow : S - 1;  // clamped; masked later
+    int krow = kt + srow < S ? (int)kt + srow : (int)S - 1;
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
       kreg[half] = *reinterpret_cast<const bf16x8_t*>(
@@ -138,105 +154,114 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
     bool has_next = kt + 64 < S;
     if (has_next) load_tile(kt + 64);  // overlap HBM with the MFMAs below
 
-    // ---- S tile = Q[16] x K[64]^T : four 16x16 C tiles (key quarters)
-    f32x4_t s_acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                        {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-#pragma unroll
-    for (int h = 0; h < 4; ++h) {
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
-            &Ks[cur][h * 16 + am][c * 32 + kg * 8]);
-        s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kf,
-                                                           s_acc[h], 0, 0, 0);
-      }
-    }
-    // scale + additive key mask + S-bound masking, then online softmax.
-    // Everything runs in the exp2 (log2) domain: v_exp_f32 IS exp2, so
-    // exp2f(x) is one instruction where __expf costs an extra multiply.
-    float sv[4][4];  // [h][r]
+    float mv2[4], oob[4];
 #pragma unroll
     for (int h = 0; h < 4; ++h) {
       int key = (int)kt + h * 16 + am;
-      float mv = (m_p && key < (int)S) ? m_p[key] * ATTN_LOG2E : 0.f;
-      float oob = key < (int)S ? 0.f : -1e30f;  // partial last tile
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        sv[h][r] = s_acc[h][r] * scale2 + mv + oob;
+      mv2[h] = (m_p && key < (int)S) ? m_p[key] * ATTN_LOG2E : 0.f;
+      oob[h] = key < (int)S ? 0.f : -1e30f;  // partial last tile
     }
-    float alpha[4];
+
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float smax = fmaxf(fmaxf(sv[0][r], sv[1][r]),
-                         fmaxf(sv[2][r], sv[3][r]));
-      float tmax = dpp16_max(smax);
-      float m_new = fmaxf(m_run[r], tmax);
-      alpha[r] = exp2f(m_run[r] - m_new);
-      float psum = 0.f;
+    for (int rb = 0; rb < RB; ++rb) {
+      // ---- S tile = Q[16] x K[64]^T : four 16x16 C tiles (key quarters)
+      f32x4_t s_acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                          {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
       for (int h = 0; h < 4; ++h) {
-        sv[h][r] = exp2f(sv[h][r] - m_new);
-        psum += sv[h][r];
-      }
-      l_run[r] = l_run[r] * alpha[r] + dpp16_sum(psum);
-      m_run[r] = m_new;
-      if (do_drop) {
-        unsigned int qrow = (unsigned int)(q0 + kg * 4 + r);
 #pragma unroll
-        for (int h = 0; h < 4; ++h) {
-          if (drop_hash(seed, (unsigned int)bh, qrow,
-                        (unsigned int)(kt + h * 16 + am)) < thresh)
-            sv[h][r] = 0.f;
+        for (int c = 0; c < 2; ++c) {
+          bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+              &Ks[cur][h * 16 + am][c * 32 + kg * 8]);
+          s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[rb][c], kf, s_acc[h], 0, 0, 0);
         }
       }
-    }
+      // online softmax in the exp2 domain (v_exp_f32 IS exp2)
+      float sv[4][4];  // [h][r]
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+      for (int h = 0; h < 4; ++h) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha[r];
-    }
-    // ---- stage (dropped) P bf16 through per-wave LDS: C-write, A-read
-#pragma unroll
-    for (int h = 0; h < 4; ++h) {
+        for (int r = 0; r < 4; ++r)
+          sv[h][r] = s_acc[h][r] * scale2 + mv2[h] + oob[h];
+      }
+      float alpha[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        __hip_bfloat16 pb = __float2bfloat16(sv[h][r]);
-        Pw[w][kg * 4 + r][h * 16 + am] = reinterpret_cast<short&>(pb);
+        float smax = fmaxf(fmaxf(sv[0][r], sv[1][r]),
+                           fmaxf(sv[2][r], sv[3][r]));
+        float tmax = dpp16_max(smax);
+        float m_new = fmaxf(m_run[rb][r], tmax);
+        alpha[r] = exp2f(m_run[rb][r] - m_new);
+        float psum = 0.f;
+#pragma unroll
+        for (int h = 0; h < 4; ++h) {
+          sv[h][r] = exp2f(sv[h][r] - m_new);
+          psum += sv[h][r];
+        }
+        l_run[rb][r] = l_run[rb][r] * alpha[r] + dpp16_sum(psum);
+        m_run[rb][r] = m_new;
+        if (do_drop) {
+          unsigned int qrow = (unsigned int)(q0 + rb * 16 + kg * 4 + r);
+#pragma unroll
+          for (int h = 0; h < 4; ++h) {
+            if (drop_hash(seed, (unsigned int)bh, qrow,
+                          (unsigned int)(kt + h * 16 + am)) < thresh)
+              sv[h][r] = 0.f;
+          }
+        }
       }
-    }
-    wave_lds_fence();
-    bf16x8_t pf[2];
 #pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      pf[kc] = *reinterpret_cast<const bf16x8_t*>(
-          &Pw[w][am][kc * 32 + kg * 8]);
-    }
-    // ---- O += P @ V : per 16-col d tile, two k-chunks of 32 keys
+      for (int dt = 0; dt < 4; ++dt) {
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+        for (int r = 0; r < 4; ++r) o_acc[rb][dt][r] *= alpha[r];
+      }
+      // ---- stage (dropped) P bf16 through per-wave LDS: C-write, A-read
+#pragma unroll
+      for (int h = 0; h < 4; ++h) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          __hip_bfloat16 pb = __float2bfloat16(sv[h][r]);
+          Pw[w][kg * 4 + r][h * 16 + am] = reinterpret_cast<short&>(pb);
+        }
+      }
+      wave_lds_fence();
+      bf16x8_t pf[2];
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
-        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-            &VsT[cur][dt * 16 + am][kc * 32 + kg * 8]);
-        o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            pf[kc], vf, o_acc[dt], 0, 0, 0);
+        pf[kc] = *reinterpret_cast<const bf16x8_t*>(
+            &Pw[w][am][kc * 32 + kg * 8]);
       }
+      // ---- O += P @ V : per 16-col d tile, two k-chunks of 32 keys
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+              &VsT[cur][dt * 16 + am][kc * 32 + kg * 8]);
+          o_acc[rb][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              pf[kc], vf, o_acc[rb][dt], 0, 0, 0);
+        }
+      }
+      wave_lds_fence();  // Pw reused by the next row block / iteration
     }
-    wave_lds_fence();  // Pw reused next iteration (wave-private)
     if (has_next) store_tile(1 - cur);
     cur = 1 - cur;
   }
   // ---- epilogue: normalize (+ dropout keep-rescale) + store
   float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt) {
+  for (int rb = 0; rb < RB; ++rb) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      long qrow = q0 + kg * 4 + r;
-      if (qrow >= S) continue;
-      float val = o_acc[dt][r] / l_run[r] * rkeep;
-      __hip_bfloat16 ob = __float2bfloat16(val);
-      o_p[qrow * ATTN_D + dt * 16 + am] = reinterpret_cast<short&>(ob);
+    for (int dt = 0; dt < 4; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long qrow = q0 + rb * 16 + kg * 4 + r;
+        if (qrow >= S) continue;
+        float val = o_acc[rb][dt][r] / l_run[rb][r] * rkeep;
+        __hip_bfloat16 ob = __float2bfloat16(val);
+        o_p[qrow * ATTN_D + dt * 16 + am] = reinterpret_cast<short&>(ob);
+      }
     }
   }
 }

@@ -439,14 +439,31 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(S % 32 == 0 && S >= 32, "S must be a multiple of 32");
   TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes());
   auto o = torch::empty_like(q);
-  dim3 grid((S + 63) / 64, B * H);
-  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, cur_stream(),
-                     reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
-                     reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
-                     reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
-                     reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
-                     attn_mask_ptr(mask, B, S), S, H, (float)scale,
-                     (float)p_drop, (unsigned int)(uint64_t)seed);
+  // RB=2 (128-row blocks) halves K/V re-read traffic for long sequences;
+  // RB=1 keeps more blocks in flight for short ones. 1-D grid with bh as
+  // the fast dimension = XCD-locality swizzle (see kernel comment).
+  long nbh = B * H;
+  if (S >= 256) {
+    long nqb = (S + 127) / 128;
+    hipLaunchKernelGGL(attn_fwd_kernel<2>, dim3(nqb * nbh), dim3(256), 0,
+                       cur_stream(),
+                       reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
+                       attn_mask_ptr(mask, B, S), S, H, nbh, (float)scale,
+                       (float)p_drop, (unsigned int)(uint64_t)seed);
+  } else {
+    long nqb = (S + 63) / 64;
+    hipLaunchKernelGGL(attn_fwd_kernel<1>, dim3(nqb * nbh), dim3(256), 0,
+                       cur_stream(),
+                       reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
+                       attn_mask_ptr(mask, B, S), S, H, nbh, (float)scale,
+                       (float)p_drop, (unsigned int)(uint64_t)seed);
+  }
   return o;
 }
 
