@@ -6,26 +6,33 @@
 // the forward's mask, no S x S state):
 //
 //   delta_i = rowsum(dO_i * O_i)
-//   dV = P^T dO          dP = dO V^T
+//   dV = D(P)^T dO       dP = mask/(1-p) o (dO V^T)
 //   dS = P o (dP - delta)            (row-wise subtract)
 //   dQ = dS K * scale    dK = dS^T Q * scale
 //
-// Two kernels, no atomics:
-//   K1 (per q-tile):   recompute per-row m/l (pass A), write M/L/delta to
-//                      global, then dQ (pass B).
-//   K2 (per key-tile): mirror structure with swapped roles — S' = K Q^T
-//                      gives C tiles [key][q]; per-column softmax stats are
-//                      read from the M/L arrays K1 wrote; accumulates dV and
-//                      dK.
-// Every MFMA fragment pattern below reuses the GPU-verified forward
-// patterns (QK^T loader, LDS C-layout->A-layout staging for P, V-style
-// B-fragments); see tests/test_mfma_probe.py for the layout contract.
+// Two kernels, no atomics, both 4-wave (256-thread) blocks that stage the
+// streamed operand tiles in LDS ONCE per block (the round-1 one-wave
+// version re-read K/V from L2 per 16 rows and was K/V-bandwidth-bound —
+// same diagnosis as the forward, fixed the same way):
+//   K1 (per 64-query block): pass A recomputes per-row m/l (K tiles via
+//      LDS), writes M/L/delta, then pass B accumulates dQ (K, V, and
+//      K-transposed tiles via LDS).
+//   K2 (per 64-key block): mirror structure with swapped roles — S' =
+//      K Q^T gives C tiles [key][q]; per-column softmax stats are read
+//      from the M/L arrays K1 wrote; accumulates dV and dK (Q, dO and
+//      their transposes via LDS).
+// Grids are 1-D with (b,h) as the fast dimension: blocks that re-read
+// the same tensors land on the same XCD's L2 (see attention.hip).
+// M/L stats are in the exp2 (log2) domain; reductions are DPP row_ror
+// (common.h). Every MFMA fragment pattern reuses the GPU-verified
+// forward patterns; see tests/test_mfma_probe.py for the layout contract.
 #include "common.h"
 
 typedef short bwd_bf16x8 __attribute__((ext_vector_type(8)));
 typedef float bwd_f32x4 __attribute__((ext_vector_type(4)));
 
 #define ATTN_BD 64
+#define BKPAD 8
 
 __device__ __forceinline__ unsigned int bwd_drop_hash(unsigned int seed,
                                                       unsigned int bh,
@@ -39,37 +46,35 @@ __device__ __forceinline__ unsigned int bwd_drop_hash(unsigned int seed,
   return x;
 }
 
-// 16-lane reductions via DPP (common.h) — no LDS-pipe traffic
-#define bwd_red_max dpp16_max
-#define bwd_red_sum dpp16_sum
-
-// load an A/B fragment row-block: elem j from src[(row)*64 + c*32 + kg*8+j]
-__device__ __forceinline__ bwd_bf16x8 frag_rowmajor(const short* src,
-                                                    long row, int c, int kg) {
-  bwd_bf16x8 f;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) f[j] = src[row * ATTN_BD + c * 32 + kg * 8 + j];
-  return f;
+// load an A/B fragment row-block from LDS rows [64][64+BKPAD]
+__device__ __forceinline__ bwd_bf16x8 lds_frag(
+    const short (*buf)[ATTN_BD + BKPAD], int row, int col) {
+  return *reinterpret_cast<const bwd_bf16x8*>(&buf[row][col]);
 }
 
 // ---------------------------------------------------------------- K1
-// grid (S/16, B*H), 64 threads. Writes M, L, delta [B*H, S] fp32 and dQ.
-__global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
-                                  const __hip_bfloat16* __restrict__ K,
-                                  const __hip_bfloat16* __restrict__ V,
-                                  const __hip_bfloat16* __restrict__ O,
-                                  const __hip_bfloat16* __restrict__ dO,
-                                  __hip_bfloat16* __restrict__ dQ,
-                                  float* __restrict__ Mbuf,
-                                  float* __restrict__ Lbuf,
-                                  float* __restrict__ Dbuf,
-                                  const float* __restrict__ mask, long S,
-                                  long H, float scale, float p_drop,
-                                  unsigned int seed) {
-  __shared__ float PS[16][32 + 1];
-  int l = threadIdx.x;
-  long bh = blockIdx.y;
-  long q0 = (long)blockIdx.x * 16;
+// grid 1-D (ceil(S/64) * B*H), 256 threads (4 waves x 16 q rows).
+// Writes M, L (exp2-domain), delta [B*H, S] fp32 and dQ.
+__global__ void
+__launch_bounds__(256, 2)
+attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
+                  const __hip_bfloat16* __restrict__ K,
+                  const __hip_bfloat16* __restrict__ V,
+                  const __hip_bfloat16* __restrict__ O,
+                  const __hip_bfloat16* __restrict__ dO,
+                  __hip_bfloat16* __restrict__ dQ,
+                  float* __restrict__ Mbuf, float* __restrict__ Lbuf,
+                  float* __restrict__ Dbuf,
+                  const float* __restrict__ mask, long S, long H, long NBH,
+                  float scale, float p_drop, unsigned int seed) {
+  __shared__ short Ks[64][ATTN_BD + BKPAD];
+  __shared__ short Vs[64][ATTN_BD + BKPAD];
+  __shared__ short KsT[ATTN_BD][64 + BKPAD];
+  __shared__ short DSw[4][16][64 + BKPAD];  // per-wave bf16 dS staging
+  int t = threadIdx.x;
+  int w = t >> 6, l = t & 63;
+  long bh = (long)blockIdx.x % NBH;
+  long q0 = ((long)blockIdx.x / NBH) * 64 + w * 16;
   const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * ATTN_BD;
   const short* k_p = reinterpret_cast<const short*>(K) + bh * S * ATTN_BD;
   const short* v_p = reinterpret_cast<const short*>(V) + bh * S * ATTN_BD;
@@ -82,41 +87,82 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
   const bool do_drop = p_drop > 0.0f;
   const float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
-  const float scale2 = scale * ATTN_LOG2E;  // Mbuf/Lbuf stats are in the
-                                            // exp2 (log2) domain
+  const float scale2 = scale * ATTN_LOG2E;  // exp2-domain logit scale
 
+  int qrow_a = q0 + am < S ? (int)q0 + am : (int)S - 1;
   bwd_bf16x8 qf[2], dof[2];
 #pragma unroll
   for (int c = 0; c < 2; ++c) {
-    qf[c] = frag_rowmajor(q_p, q0 + am, c, kg);
-    dof[c] = frag_rowmajor(do_p, q0 + am, c, kg);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      qf[c][j] = q_p[qrow_a * ATTN_BD + c * 32 + kg * 8 + j];
+      dof[c][j] = do_p[qrow_a * ATTN_BD + c * 32 + kg * 8 + j];
+    }
   }
 
-  // ---- pass A: softmax stats m, l per row
+  int srow = t >> 2, scol = (t & 3) * 16;
+  auto stage_k = [&](long kt) {
+    int krow = kt + srow < S ? (int)kt + srow : (int)S - 1;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      *reinterpret_cast<bwd_bf16x8*>(&Ks[srow][scol + half * 8]) =
+          *reinterpret_cast<const bwd_bf16x8*>(
+              &k_p[krow * ATTN_BD + scol + half * 8]);
+    }
+  };
+  auto stage_kvt = [&](long kt) {
+    int krow = kt + srow < S ? (int)kt + srow : (int)S - 1;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      bwd_bf16x8 kv = *reinterpret_cast<const bwd_bf16x8*>(
+          &k_p[krow * ATTN_BD + scol + half * 8]);
+      *reinterpret_cast<bwd_bf16x8*>(&Ks[srow][scol + half * 8]) = kv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) KsT[scol + half * 8 + j][srow] = kv[j];
+      *reinterpret_cast<bwd_bf16x8*>(&Vs[srow][scol + half * 8]) =
+          *reinterpret_cast<const bwd_bf16x8*>(
+              &v_p[krow * ATTN_BD + scol + half * 8]);
+    }
+  };
+
+  // ---- pass A: softmax stats m, l per row (exp2 domain)
   float m_run[4], l_run[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
-  for (long kt = 0; kt < S; kt += 32) {
-    bwd_f32x4 s_acc[2] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
+  for (long kt = 0; kt < S; kt += 64) {
+    __syncthreads();
+    stage_k(kt);
+    __syncthreads();
+    bwd_f32x4 s_acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0},
+                          {0, 0, 0, 0}, {0, 0, 0, 0}};
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int h = 0; h < 4; ++h) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        bwd_bf16x8 kf = frag_rowmajor(k_p, kt + h * 16 + am, c, kg);
+        bwd_bf16x8 kf = lds_frag(Ks, h * 16 + am, c * 32 + kg * 8);
         s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kf,
                                                            s_acc[h], 0, 0, 0);
       }
     }
-    float mv0 = m_p ? m_p[kt + am] * ATTN_LOG2E : 0.f;
-    float mv1 = m_p ? m_p[kt + 16 + am] * ATTN_LOG2E : 0.f;
+    float mv2[4], oob[4];
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
+      int key = (int)kt + h * 16 + am;
+      mv2[h] = (m_p && key < (int)S) ? m_p[key] * ATTN_LOG2E : 0.f;
+      oob[h] = key < (int)S ? 0.f : -1e30f;
+    }
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float s0 = s_acc[0][r] * scale2 + mv0;
-      float s1 = s_acc[1][r] * scale2 + mv1;
-      float tmax = bwd_red_max(fmaxf(s0, s1));
+      float sv[4];
+#pragma unroll
+      for (int h = 0; h < 4; ++h)
+        sv[h] = s_acc[h][r] * scale2 + mv2[h] + oob[h];
+      float tmax = dpp16_max(fmaxf(fmaxf(sv[0], sv[1]),
+                                   fmaxf(sv[2], sv[3])));
       float m_new = fmaxf(m_run[r], tmax);
       float alpha = exp2f(m_run[r] - m_new);
-      float rsum = bwd_red_sum(exp2f(s0 - m_new) + exp2f(s1 - m_new));
+      float rsum = dpp16_sum(exp2f(sv[0] - m_new) + exp2f(sv[1] - m_new) +
+                             exp2f(sv[2] - m_new) + exp2f(sv[3] - m_new));
       l_run[r] = l_run[r] * alpha + rsum;
       m_run[r] = m_new;
     }
@@ -126,19 +172,19 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     long row = q0 + kg * 4 + r;
+    long rowc = row < S ? row : S - 1;
     float part = 0.f;
-    // each of the 16 lanes in the group sums 4 d-columns: d = am*4..am*4+3
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
       int d = am * 4 + u;
       float ov = __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
-          o_p)[row * ATTN_BD + d]);
+          o_p)[rowc * ATTN_BD + d]);
       float dv = __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
-          do_p)[row * ATTN_BD + d]);
+          do_p)[rowc * ATTN_BD + d]);
       part += ov * dv;
     }
-    delta[r] = bwd_red_sum(part);
-    if (am == 0) {  // one lane per (kg, r) row writes the stats
+    delta[r] = dpp16_sum(part);
+    if (am == 0 && row < S) {  // one lane per (kg, r) row writes the stats
       Mbuf[bh * S + row] = m_run[r];
       Lbuf[bh * S + row] = l_run[r];
       Dbuf[bh * S + row] = delta[r];
@@ -148,16 +194,21 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
   bwd_f32x4 dq_acc[4];
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt) dq_acc[dt] = {0, 0, 0, 0};
-  for (long kt = 0; kt < S; kt += 32) {
+  for (long kt = 0; kt < S; kt += 64) {
+    __syncthreads();
+    stage_kvt(kt);
+    __syncthreads();
     // S tile and dP tile (dP = dO V^T: same shape as QK^T with Q->dO, K->V)
-    bwd_f32x4 s_acc[2] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
-    bwd_f32x4 dp_acc[2] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
+    bwd_f32x4 s_acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0},
+                          {0, 0, 0, 0}, {0, 0, 0, 0}};
+    bwd_f32x4 dp_acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0},
+                           {0, 0, 0, 0}, {0, 0, 0, 0}};
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int h = 0; h < 4; ++h) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        bwd_bf16x8 kf = frag_rowmajor(k_p, kt + h * 16 + am, c, kg);
-        bwd_bf16x8 vf = frag_rowmajor(v_p, kt + h * 16 + am, c, kg);
+        bwd_bf16x8 kf = lds_frag(Ks, h * 16 + am, c * 32 + kg * 8);
+        bwd_bf16x8 vf = lds_frag(Vs, h * 16 + am, c * 32 + kg * 8);
         s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kf,
                                                            s_acc[h], 0, 0, 0);
         dp_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof[c], vf,
@@ -166,47 +217,51 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
     }
     // dS = P * (keep*dP/(1-p) - delta) * scale (fold dQ's trailing scale)
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      float mv = m_p ? m_p[kt + h * 16 + am] * ATTN_LOG2E : 0.f;
+    for (int h = 0; h < 4; ++h) {
+      int key = (int)kt + h * 16 + am;
+      float mv2 = (m_p && key < (int)S) ? m_p[key] * ATTN_LOG2E : 0.f;
+      float oob = key < (int)S ? 0.f : -1e30f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = exp2f(s_acc[h][r] * scale2 + mv - m_run[r]) / l_run[r];
+        float p = exp2f(s_acc[h][r] * scale2 + mv2 + oob - m_run[r])
+                  / l_run[r];
         float dp = dp_acc[h][r];
         if (do_drop) {
           unsigned int keep = bwd_drop_hash(
               seed, (unsigned int)bh, (unsigned int)(q0 + kg * 4 + r),
-              (unsigned int)(kt + h * 16 + am)) >= thresh;
+              (unsigned int)key) >= thresh;
           dp = keep ? dp * rkeep : 0.f;
         }
         float ds = p * (dp - delta[r]) * scale;
-        PS[kg * 4 + r][h * 16 + am] = ds;
+        __hip_bfloat16 b = __float2bfloat16(ds);
+        DSw[w][kg * 4 + r][h * 16 + am] = reinterpret_cast<short&>(b);
       }
     }
-    __syncthreads();
-    bwd_bf16x8 dsf;
+    wave_lds_fence();
+    bwd_bf16x8 dsf[2];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      __hip_bfloat16 b = __float2bfloat16(PS[am][kg * 8 + j]);
-      dsf[j] = reinterpret_cast<short&>(b);
-    }
-    // dQ += dS K : m=q, n=d(16/tile), k=key(32)
+    for (int kc = 0; kc < 2; ++kc)
+      dsf[kc] = *reinterpret_cast<const bwd_bf16x8*>(
+          &DSw[w][am][kc * 32 + kg * 8]);
+    // dQ += dS K : m=q, n=d(16/tile), k=keys(64 in 2 chunks)
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
-      bwd_bf16x8 kf;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        kf[j] = k_p[(kt + kg * 8 + j) * ATTN_BD + dt * 16 + am];
+      for (int kc = 0; kc < 2; ++kc) {
+        bwd_bf16x8 kfT = *reinterpret_cast<const bwd_bf16x8*>(
+            &KsT[dt * 16 + am][kc * 32 + kg * 8]);
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf[kc], kfT, dq_acc[dt], 0, 0, 0);
       }
-      dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, kf,
-                                                           dq_acc[dt], 0, 0, 0);
     }
-    __syncthreads();
+    wave_lds_fence();
   }
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       long row = q0 + kg * 4 + r;
+      if (row >= S) continue;
       __hip_bfloat16 b = __float2bfloat16(dq_acc[dt][r]);
       dq_p[row * ATTN_BD + dt * 16 + am] = reinterpret_cast<short&>(b);
     }
@@ -214,24 +269,31 @@ __global__ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
 }
 
 // ---------------------------------------------------------------- K2
-// grid (S/16, B*H): per KEY tile, accumulate dK and dV over all q.
-__global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
-                                   const __hip_bfloat16* __restrict__ K,
-                                   const __hip_bfloat16* __restrict__ V,
-                                   const __hip_bfloat16* __restrict__ dO,
-                                   __hip_bfloat16* __restrict__ dK,
-                                   __hip_bfloat16* __restrict__ dV,
-                                   const float* __restrict__ Mbuf,
-                                   const float* __restrict__ Lbuf,
-                                   const float* __restrict__ Dbuf,
-                                   const float* __restrict__ mask, long S,
-                                   long H, float scale, float p_drop,
-                                   unsigned int seed) {
-  __shared__ float PS[16][32 + 1];   // P' or dS' tile [key][q-chunk]
-  __shared__ float PS2[16][32 + 1];
-  int l = threadIdx.x;
-  long bh = blockIdx.y;
-  long k0 = (long)blockIdx.x * 16;   // this block's 16 keys
+// grid 1-D (ceil(S/64) * B*H), 256 threads (4 waves x 16 keys): per
+// 64-KEY block, accumulate dK and dV over all q (Q/dO tiles via LDS).
+__global__ void
+__launch_bounds__(256, 2)
+attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
+                   const __hip_bfloat16* __restrict__ K,
+                   const __hip_bfloat16* __restrict__ V,
+                   const __hip_bfloat16* __restrict__ dO,
+                   __hip_bfloat16* __restrict__ dK,
+                   __hip_bfloat16* __restrict__ dV,
+                   const float* __restrict__ Mbuf,
+                   const float* __restrict__ Lbuf,
+                   const float* __restrict__ Dbuf,
+                   const float* __restrict__ mask, long S, long H, long NBH,
+                   float scale, float p_drop, unsigned int seed) {
+  __shared__ short Qs[64][ATTN_BD + BKPAD];
+  __shared__ short dOs[64][ATTN_BD + BKPAD];
+  __shared__ short QsT[ATTN_BD][64 + BKPAD];
+  __shared__ short dOsT[ATTN_BD][64 + BKPAD];
+  __shared__ short Pw[4][16][64 + BKPAD];   // P' staging (bf16)
+  __shared__ short DSw[4][16][64 + BKPAD];  // dS' staging (bf16)
+  int t = threadIdx.x;
+  int w = t >> 6, l = t & 63;
+  long bh = (long)blockIdx.x % NBH;
+  long k0 = ((long)blockIdx.x / NBH) * 64 + w * 16;  // this wave's 16 keys
   const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * ATTN_BD;
   const short* k_p = reinterpret_cast<const short*>(K) + bh * S * ATTN_BD;
   const short* v_p = reinterpret_cast<const short*>(V) + bh * S * ATTN_BD;
@@ -244,48 +306,84 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
   const bool do_drop = p_drop > 0.0f;
   const float rkeep = do_drop ? 1.0f / (1.0f - p_drop) : 1.0f;
-  const float scale2 = scale * ATTN_LOG2E;  // Mbuf/Lbuf stats are in the
-                                            // exp2 (log2) domain
+  const float scale2 = scale * ATTN_LOG2E;
 
+  int krow_a = k0 + am < S ? (int)k0 + am : (int)S - 1;
   bwd_bf16x8 kf[2], vf[2];
 #pragma unroll
   for (int c = 0; c < 2; ++c) {
-    kf[c] = frag_rowmajor(k_p, k0 + am, c, kg);
-    vf[c] = frag_rowmajor(v_p, k0 + am, c, kg);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      kf[c][j] = k_p[krow_a * ATTN_BD + c * 32 + kg * 8 + j];
+      vf[c][j] = v_p[krow_a * ATTN_BD + c * 32 + kg * 8 + j];
+    }
   }
+  // additive mask value of this lane's OWN key rows (kg*4+r)
+  float mvk2[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    long key = k0 + kg * 4 + r;
+    mvk2[r] = (m_p && key < S) ? m_p[key] * ATTN_LOG2E : 0.f;
+  }
+
   bwd_f32x4 dk_acc[4], dv_acc[4];
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt) {
     dk_acc[dt] = {0, 0, 0, 0};
     dv_acc[dt] = {0, 0, 0, 0};
   }
-  for (long qt = 0; qt < S; qt += 32) {
-    // S' = K Q^T and dP' = V dO^T : C tiles [key][q] (two q halves)
-    bwd_f32x4 s_acc[2] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
-    bwd_f32x4 dp_acc[2] = {{0, 0, 0, 0}, {0, 0, 0, 0}};
+
+  int srow = t >> 2, scol = (t & 3) * 16;
+  auto stage_qdo = [&](long qt) {
+    int qrow = qt + srow < S ? (int)qt + srow : (int)S - 1;
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
+    for (int half = 0; half < 2; ++half) {
+      bwd_bf16x8 qv = *reinterpret_cast<const bwd_bf16x8*>(
+          &q_p[qrow * ATTN_BD + scol + half * 8]);
+      *reinterpret_cast<bwd_bf16x8*>(&Qs[srow][scol + half * 8]) = qv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) QsT[scol + half * 8 + j][srow] = qv[j];
+      bwd_bf16x8 dv8 = *reinterpret_cast<const bwd_bf16x8*>(
+          &do_p[qrow * ATTN_BD + scol + half * 8]);
+      *reinterpret_cast<bwd_bf16x8*>(&dOs[srow][scol + half * 8]) = dv8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dOsT[scol + half * 8 + j][srow] = dv8[j];
+    }
+  };
+
+  for (long qt = 0; qt < S; qt += 64) {
+    __syncthreads();
+    stage_qdo(qt);
+    __syncthreads();
+    // S' = K Q^T and dP' = V dO^T : C tiles [key][q] (four q quarters)
+    bwd_f32x4 s_acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0},
+                          {0, 0, 0, 0}, {0, 0, 0, 0}};
+    bwd_f32x4 dp_acc[4] = {{0, 0, 0, 0}, {0, 0, 0, 0},
+                           {0, 0, 0, 0}, {0, 0, 0, 0}};
+#pragma unroll
+    for (int h = 0; h < 4; ++h) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        bwd_bf16x8 qf = frag_rowmajor(q_p, qt + h * 16 + am, c, kg);
-        bwd_bf16x8 dof = frag_rowmajor(do_p, qt + h * 16 + am, c, kg);
-        s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[c], qf,
+        bwd_bf16x8 qfr = lds_frag(Qs, h * 16 + am, c * 32 + kg * 8);
+        bwd_bf16x8 dofr = lds_frag(dOs, h * 16 + am, c * 32 + kg * 8);
+        s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[c], qfr,
                                                            s_acc[h], 0, 0, 0);
-        dp_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[c], dof,
+        dp_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[c], dofr,
                                                             dp_acc[h], 0, 0, 0);
       }
     }
     // per-column (q) stats; column index = h*16 + am
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      long qrow = qt + h * 16 + am;
-      float m_q = Mbuf[bh * S + qrow];
-      float l_q = Lbuf[bh * S + qrow];
-      float d_q = Dbuf[bh * S + qrow];
+    for (int h = 0; h < 4; ++h) {
+      int qrow = (int)qt + h * 16 + am;
+      int qrc = qrow < (int)S ? qrow : (int)S - 1;
+      float m_q = Mbuf[bh * S + qrc];
+      float l_q = Lbuf[bh * S + qrc];
+      float d_q = Dbuf[bh * S + qrc];
+      float oob = qrow < (int)S ? 0.f : -1e30f;  // oob q contributes 0
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float mv = m_p ? m_p[k0 + kg * 4 + r] * ATTN_LOG2E : 0.f;  // key mask
-        float p = exp2f(s_acc[h][r] * scale2 + mv - m_q) / l_q;
+        float p = exp2f(s_acc[h][r] * scale2 + mvk2[r] + oob - m_q) / l_q;
         float pd = p, dp = dp_acc[h][r];
         if (do_drop) {
           unsigned int keep = bwd_drop_hash(
@@ -294,41 +392,45 @@ __global__ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
           pd = keep ? p * rkeep : 0.f;   // dropped P' for dV
           dp = keep ? dp * rkeep : 0.f;  // dropped dP' for dK
         }
-        PS[kg * 4 + r][h * 16 + am] = pd;                       // P'
-        PS2[kg * 4 + r][h * 16 + am] = p * (dp - d_q) * scale;
+        float ds = p * (dp - d_q) * scale;
+        __hip_bfloat16 b1 = __float2bfloat16(pd);
+        __hip_bfloat16 b2 = __float2bfloat16(ds);
+        Pw[w][kg * 4 + r][h * 16 + am] = reinterpret_cast<short&>(b1);
+        DSw[w][kg * 4 + r][h * 16 + am] = reinterpret_cast<short&>(b2);
       }
     }
-    __syncthreads();
-    bwd_bf16x8 pf, dsf;
+    wave_lds_fence();
+    bwd_bf16x8 pf[2], dsf[2];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      __hip_bfloat16 b1 = __float2bfloat16(PS[am][kg * 8 + j]);
-      __hip_bfloat16 b2 = __float2bfloat16(PS2[am][kg * 8 + j]);
-      pf[j] = reinterpret_cast<short&>(b1);
-      dsf[j] = reinterpret_cast<short&>(b2);
+    for (int kc = 0; kc < 2; ++kc) {
+      pf[kc] = *reinterpret_cast<const bwd_bf16x8*>(
+          &Pw[w][am][kc * 32 + kg * 8]);
+      dsf[kc] = *reinterpret_cast<const bwd_bf16x8*>(
+          &DSw[w][am][kc * 32 + kg * 8]);
     }
     // dV += P' dO : m=key, n=d, k=q ; dK += dS' Q : m=key, n=d, k=q
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
-      bwd_bf16x8 dof, qf;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        long qq = qt + kg * 8 + j;
-        dof[j] = do_p[qq * ATTN_BD + dt * 16 + am];
-        qf[j] = q_p[qq * ATTN_BD + dt * 16 + am];
+      for (int kc = 0; kc < 2; ++kc) {
+        bwd_bf16x8 dofT = *reinterpret_cast<const bwd_bf16x8*>(
+            &dOsT[dt * 16 + am][kc * 32 + kg * 8]);
+        bwd_bf16x8 qfT = *reinterpret_cast<const bwd_bf16x8*>(
+            &QsT[dt * 16 + am][kc * 32 + kg * 8]);
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pf[kc], dofT, dv_acc[dt], 0, 0, 0);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsf[kc], qfT, dk_acc[dt], 0, 0, 0);
       }
-      dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof,
-                                                           dv_acc[dt], 0, 0, 0);
-      dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf,
-                                                           dk_acc[dt], 0, 0, 0);
     }
-    __syncthreads();
+    wave_lds_fence();
   }
 #pragma unroll
   for (int dt = 0; dt < 4; ++dt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       long row = k0 + kg * 4 + r;
+      if (row >= S) continue;
       __hip_bfloat16 bk = __float2bfloat16(dk_acc[dt][r]);
       __hip_bfloat16 bv = __float2bfloat16(dv_acc[dt][r]);
       dk_p[row * ATTN_BD + dt * 16 + am] = reinterpret_cast<short&>(bk);

@@ -496,18 +496,20 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto Dbuf = torch::empty({B * H * S}, fopt);
   const float* mp = attn_mask_ptr(mask, B, S);
   unsigned int sd = (unsigned int)(uint64_t)seed;
-  dim3 grid(S / 16, B * H);
+  // 1-D grids, (b,h) fast for XCD L2 locality (see attention.hip)
+  long nbh = B * H;
+  dim3 grid(((S + 63) / 64) * nbh);
   auto st = cur_stream();
 #define BF16P(t) reinterpret_cast<__hip_bfloat16*>((t).data_ptr())
-  hipLaunchKernelGGL(attn_bwd_q_kernel, grid, dim3(64), 0, st, BF16P(q),
+  hipLaunchKernelGGL(attn_bwd_q_kernel, grid, dim3(256), 0, st, BF16P(q),
                      BF16P(k), BF16P(v), BF16P(o), BF16P(dout), BF16P(dq),
                      Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
-                     Dbuf.data_ptr<float>(), mp, S, H, (float)scale,
+                     Dbuf.data_ptr<float>(), mp, S, H, nbh, (float)scale,
                      (float)p_drop, sd);
-  hipLaunchKernelGGL(attn_bwd_kv_kernel, grid, dim3(64), 0, st, BF16P(q),
+  hipLaunchKernelGGL(attn_bwd_kv_kernel, grid, dim3(256), 0, st, BF16P(q),
                      BF16P(k), BF16P(v), BF16P(dout), BF16P(dk), BF16P(dv),
                      Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
-                     Dbuf.data_ptr<float>(), mp, S, H, (float)scale,
+                     Dbuf.data_ptr<float>(), mp, S, H, nbh, (float)scale,
                      (float)p_drop, sd);
 #undef BF16P
   return {dq, dk, dv};
