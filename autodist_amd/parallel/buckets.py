@@ -120,10 +120,15 @@ class Bucket:
             ev = torch.cuda.Event()
             ev.record()  # grads complete on the compute (current) stream
             engine.comm_stream.wait_event(ev)
+            t0 = None
+            if engine.comm_sampling:
+                t0 = torch.cuda.Event(enable_timing=True)
+                t0.record(engine.comm_stream)
             with torch.cuda.stream(engine.comm_stream):
                 self._reduce(engine)
             self.done_event = torch.cuda.Event()
             self.done_event.record(engine.comm_stream)
+            self._t0 = t0
         else:
             self._reduce(engine)
 
@@ -139,9 +144,14 @@ class Bucket:
         if engine.device.type == "cuda":
             with torch.cuda.stream(engine.comm_stream):
                 self.compressor.finalize(self.flat, self._handle)
-            self.done_event = torch.cuda.Event()
+            self.done_event = torch.cuda.Event(
+                enable_timing=engine.comm_sampling)
             self.done_event.record(engine.comm_stream)
             torch.cuda.current_stream().wait_event(self.done_event)
+            if engine.comm_sampling and getattr(self, "_t0", None) is not None:
+                engine.record_comm_sample(self.nbytes, self._t0,
+                                          self.done_event)
+                self._t0 = None
         else:
             self.compressor.finalize(self.flat, self._handle)
 

@@ -199,9 +199,38 @@ class DistributedEngine:
         # single-GPU proof available.
         self._force_collectives = os.environ.get(
             "AUTODIST_FORCE_COLLECTIVES", "") in ("1", "True")
+        # AUTODIST_COMM_SAMPLES=<path>: record (nbytes, world, seconds) of
+        # every bucket all-reduce (comm-stream event timing) and dump JSON
+        # at drain() — calibration data for the simulator cost model
+        # (simulator/cost_model.py fit)
+        self._comm_sample_path = os.environ.get("AUTODIST_COMM_SAMPLES", "")
+        self._comm_events: list = []
         # note: RCCL supports ReduceOp.AVG, but the mean is instead fused as
         # a scale into the compress/cast kernels (one code path for gloo +
         # every compressor)
+
+    @property
+    def comm_sampling(self) -> bool:
+        return bool(self._comm_sample_path) and self.device.type == "cuda"
+
+    def record_comm_sample(self, nbytes, ev0, ev1):
+        self._comm_events.append((nbytes, ev0, ev1))
+
+    def flush_comm_samples(self):
+        """Resolve recorded events into (nbytes, world, seconds) samples and
+        append them to AUTODIST_COMM_SAMPLES as JSON lines."""
+        if not self._comm_events:
+            return []
+        torch.cuda.synchronize(self.device)
+        samples = [(n, self.world_size, ev0.elapsed_time(ev1) / 1e3)
+                   for n, ev0, ev1 in self._comm_events]
+        self._comm_events.clear()
+        if self._comm_sample_path:
+            import json
+            with open(self._comm_sample_path, "a") as f:
+                for s_ in samples:
+                    f.write(json.dumps(s_) + "\n")
+        return samples
 
     @property
     def collectives_active(self) -> bool:
@@ -637,6 +666,8 @@ class DistributedEngine:
         """Consume ALL outstanding PS rounds (end of training / checkpoint)."""
         for grp in self.ps_groups:
             grp.drain(self)
+        if self.comm_sampling:
+            self.flush_comm_samples()
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)
 

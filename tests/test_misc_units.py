@@ -104,3 +104,39 @@ def test_coordinator_fail_fast(tmp_path):
                        text=True, timeout=120)
     assert r.returncode != 0
     assert "SHOULD NOT REACH" not in r.stdout
+
+
+def test_cost_model_fit_latency_bandwidth():
+    """fit() recovers both terms from synthetic t = a + wire/bw samples."""
+    from autodist_amd.simulator.cost_model import CostModel
+    cm = CostModel(calibration=None)
+    lat, bw = 30e-6, 400e9
+    samples = []
+    for nbytes in (1 << 16, 1 << 20, 1 << 24, 1 << 27):
+        wire = 2.0 * 7 / 8 * nbytes
+        samples.append((nbytes, 8, lat + wire / bw))
+    cm.fit(samples)
+    assert abs(cm.coll_latency - lat) / lat < 0.05
+    got_bw = cm.allreduce_eff * cm.links * cm.link_gbps * 1e9
+    assert abs(got_bw - bw) / bw < 0.05
+
+
+def test_cost_model_calibration_roundtrip(tmp_path):
+    from autodist_amd.simulator.cost_model import CostModel
+    cm = CostModel(calibration=None)
+    cm.allreduce_eff = 0.42
+    cm.coll_latency = 33e-6
+    path = str(tmp_path / "cal.json")
+    cm.save_calibration(path, measured_on="unit test",
+                        samples=[[1024, 8, 1e-4]])
+    cm2 = CostModel(calibration=path)
+    assert cm2.allreduce_eff == 0.42
+    assert cm2.coll_latency == 33e-6
+    assert cm2.calibrated_from == "unit test"
+
+
+def test_cost_model_fit_latency_only():
+    from autodist_amd.simulator.cost_model import CostModel
+    cm = CostModel(calibration=None)
+    cm.fit_latency([(1024, 1, 25e-6), (2048, 1, 27e-6), (4096, 1, 29e-6)])
+    assert abs(cm.coll_latency - 27e-6) < 1e-9
