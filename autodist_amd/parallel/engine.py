@@ -199,10 +199,13 @@ class DistributedEngine:
         # single-GPU proof available.
         self._force_collectives = os.environ.get(
             "AUTODIST_FORCE_COLLECTIVES", "") in ("1", "True")
-        # AUTODIST_COMM_SAMPLES=<path>: record (nbytes, world, seconds) of
-        # every bucket all-reduce (comm-stream event timing) and dump JSON
-        # at drain() — calibration data for the simulator cost model
-        # (simulator/cost_model.py fit)
+        # AUTODIST_COMM_SAMPLES=<path>: record (nbytes, world, seconds)
+        # comm-stream windows of every bucket all-reduce and dump JSON at
+        # drain(). NOTE: these measure the full issue->finalize window
+        # (includes queueing behind earlier buckets while overlapped with
+        # backward) — an overlap/occupancy diagnostic. Bare-collective
+        # calibration for the cost model comes from the serialized
+        # tools/comm_microbench.py sweep instead.
         self._comm_sample_path = os.environ.get("AUTODIST_COMM_SAMPLES", "")
         self._comm_events: list = []
         # note: RCCL supports ReduceOp.AVG, but the mean is instead fused as
