@@ -51,7 +51,8 @@ class _ShardedLookup(torch.autograd.Function):
         if world <= 1:
             ctx.save_for_backward(ids)
             ctx.module = module
-            return _local_gather(shard_weight, ids)
+            return _local_gather(shard_weight, ids.reshape(-1)).view(
+                *ids.shape, module.dim)
         group = module.process_group
         flat_ids = ids.reshape(-1)
         owner = torch.bucketize(flat_ids, module.boundaries, right=True)

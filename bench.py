@@ -196,7 +196,10 @@ def build_lm1b(args, device, rank, world, use_cuda):
     strategy_name = args.strategy or "AutoStrategy"
     B = args.batch_size or 128
     S = args.seq_len or 20
-    model = lm1b_full().to(device)
+    # vocab-parallel tied softmax (parallel/vocab_parallel.py): the 793k x
+    # 512 projection+CE shards across ranks (3 tiny collectives instead of
+    # full-vocab logits); exact-match vs dense CE is test-covered
+    model = lm1b_full(sharded_softmax=True).to(device)
     opt = torch.optim.Adagrad(model.parameters(), lr=0.2)
     engine = _make_engine(model, opt, strategy_name, rank, world, device,
                           args.bucket_mb)
@@ -218,7 +221,8 @@ def build_lm1b(args, device, rank, world, use_cuda):
                 unit="words/sec", graph_safe=False,  # cuDNN-style LSTM
                 config={"model": "lm1b_lstm", "global_batch": world * B,
                         "seq_len": S, "parallelism": f"dp{world}",
-                        "strategy": strategy_name, "vocab": vocab})
+                        "strategy": strategy_name, "vocab": vocab,
+                        "softmax": "vocab_parallel"})
 
 
 BUILDERS = {"resnet50": build_resnet, "resnet101": build_resnet,
