@@ -100,6 +100,25 @@ def fused_apply(cls_name: str, param: torch.Tensor, grad: torch.Tensor,
                      cls_name == "AdamW", bc1, sqrt_bc2,
                      bool(hyper.get("maximize", False)))
         return True
+    if cls_name == "Adagrad":
+        state["step"] += 1
+        clr = float(hyper["lr"]) / (
+            1.0 + (float(state["step"]) - 1.0)
+            * float(hyper.get("lr_decay", 0.0)))
+        e.fused_adagrad(param, grad, state["sum"], clr,
+                        float(hyper.get("eps", 1e-10)),
+                        float(hyper.get("weight_decay", 0.0)),
+                        bool(hyper.get("maximize", False)))
+        return True
+    if cls_name == "RMSprop":
+        e.fused_rmsprop(param, grad, state["square_avg"],
+                        state.get("grad_avg"), state.get("momentum_buffer"),
+                        float(hyper["lr"]), float(hyper.get("alpha", 0.99)),
+                        float(hyper.get("eps", 1e-8)),
+                        float(hyper.get("weight_decay", 0.0)),
+                        float(hyper.get("momentum", 0.0)),
+                        bool(hyper.get("maximize", False)))
+        return True
     return False
 
 

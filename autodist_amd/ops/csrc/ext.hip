@@ -288,6 +288,41 @@ void fused_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      maximize);
 }
 
+void fused_adagrad(torch::Tensor p, torch::Tensor g, torch::Tensor sum,
+                   double clr, double eps, double weight_decay,
+                   bool maximize) {
+  check_f32_flat(p, "param");
+  check_f32_flat(g, "grad");
+  check_f32_flat(sum, "sum");
+  long n = p.numel();
+  hipLaunchKernelGGL(fused_adagrad_kernel, dim3(grid_for(n)),
+                     dim3(BLOCK_THREADS), 0, cur_stream(),
+                     p.data_ptr<float>(), g.data_ptr<float>(),
+                     sum.data_ptr<float>(), n, (float)clr, (float)eps,
+                     (float)weight_decay, maximize);
+}
+
+void fused_rmsprop(torch::Tensor p, torch::Tensor g, torch::Tensor sq,
+                   c10::optional<torch::Tensor> ga,
+                   c10::optional<torch::Tensor> buf, double lr, double alpha,
+                   double eps, double weight_decay, double momentum,
+                   bool maximize) {
+  check_f32_flat(p, "param");
+  check_f32_flat(g, "grad");
+  check_f32_flat(sq, "square_avg");
+  long n = p.numel();
+  float* gap = (ga.has_value() && ga->defined()) ? ga->data_ptr<float>()
+                                                 : nullptr;
+  float* bufp = (buf.has_value() && buf->defined()) ? buf->data_ptr<float>()
+                                                    : nullptr;
+  hipLaunchKernelGGL(fused_rmsprop_kernel, dim3(grid_for(n, 1)),
+                     dim3(BLOCK_THREADS), 0, cur_stream(),
+                     p.data_ptr<float>(), g.data_ptr<float>(),
+                     sq.data_ptr<float>(), gap, bufp, n, (float)lr,
+                     (float)alpha, (float)eps, (float)weight_decay,
+                     (float)momentum, maximize);
+}
+
 void scale_cast_bf16(torch::Tensor in, torch::Tensor out, double scale) {
   check_f32_flat(in, "in");
   TORCH_CHECK(out.scalar_type() == torch::kBFloat16 && out.is_contiguous());
@@ -541,6 +576,10 @@ void psgd_add_err_pad(torch::Tensor flat, torch::Tensor err,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused flat SGD update (gfx950)");
   m.def("fused_adam", &fused_adam, "fused flat Adam/AdamW update (gfx950)");
+  m.def("fused_adagrad", &fused_adagrad,
+        "fused flat Adagrad update (gfx950)");
+  m.def("fused_rmsprop", &fused_rmsprop,
+        "fused flat RMSprop update (gfx950, centered/momentum variants)");
   m.def("scale_cast_bf16", &scale_cast_bf16, "scale+cast fp32->bf16");
   m.def("cast_back_f32", &cast_back_f32, "cast bf16->fp32");
   m.def("ef_compress", &ef_compress, "fused error-feedback bf16 compress");

@@ -199,3 +199,81 @@ __global__ void gather_rows_kernel(const float* __restrict__ src,
     out[i] = src[idx[r] * dim + c];
   }
 }
+
+// ---------------------------------------------------------------- Adagrad
+// g' = (max? -g : g) + wd*p ; sum += g'^2 ; p -= clr * g'/(sqrt(sum)+eps)
+// (clr = lr / (1 + (step-1)*lr_decay), host-computed)
+__global__ void fused_adagrad_kernel(float* __restrict__ p,
+                                     const float* __restrict__ g,
+                                     float* __restrict__ sum, long n,
+                                     float clr, float eps,
+                                     float weight_decay, int maximize) {
+  long stride = (long)gridDim.x * blockDim.x * 4;
+  for (long base = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       base < n; base += stride) {
+    if (base + 3 < n) {
+      float4 pv = *reinterpret_cast<float4*>(p + base);
+      float4 gv = *reinterpret_cast<const float4*>(g + base);
+      float4 sv = *reinterpret_cast<float4*>(sum + base);
+      float pr[4] = {pv.x, pv.y, pv.z, pv.w};
+      float gr[4] = {gv.x, gv.y, gv.z, gv.w};
+      float sr[4] = {sv.x, sv.y, sv.z, sv.w};
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float d = maximize ? -gr[k] : gr[k];
+        d += weight_decay * pr[k];
+        sr[k] += d * d;
+        pr[k] -= clr * d / (sqrtf(sr[k]) + eps);
+      }
+      *reinterpret_cast<float4*>(sum + base) =
+          make_float4(sr[0], sr[1], sr[2], sr[3]);
+      *reinterpret_cast<float4*>(p + base) =
+          make_float4(pr[0], pr[1], pr[2], pr[3]);
+    } else {
+      for (long i = base; i < n; ++i) {
+        float d = maximize ? -g[i] : g[i];
+        d += weight_decay * p[i];
+        sum[i] += d * d;
+        p[i] -= clr * d / (sqrtf(sum[i]) + eps);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- RMSprop
+// g' = (max? -g : g) + wd*p ; sq = a*sq + (1-a)g'^2 ;
+// centered: ga = ga + (1-a)(g'-ga), denom = sqrt(sq - ga^2)+eps
+//           else denom = sqrt(sq)+eps
+// momentum: buf = mom*buf + g'/denom, p -= lr*buf ; else p -= lr*g'/denom
+__global__ void fused_rmsprop_kernel(float* __restrict__ p,
+                                     const float* __restrict__ g,
+                                     float* __restrict__ sq,
+                                     float* __restrict__ ga,   // nullable
+                                     float* __restrict__ buf,  // nullable
+                                     long n, float lr, float alpha,
+                                     float eps, float weight_decay,
+                                     float momentum, int maximize) {
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float d = maximize ? -g[i] : g[i];
+    d += weight_decay * p[i];
+    float s = alpha * sq[i] + (1.f - alpha) * d * d;
+    sq[i] = s;
+    float denom;
+    if (ga != nullptr) {
+      float a = ga[i] + (1.f - alpha) * (d - ga[i]);
+      ga[i] = a;
+      denom = sqrtf(s - a * a) + eps;
+    } else {
+      denom = sqrtf(s) + eps;
+    }
+    if (buf != nullptr) {
+      float b = momentum * buf[i] + d / denom;
+      buf[i] = b;
+      p[i] -= lr * b;
+    } else {
+      p[i] -= lr * d / denom;
+    }
+  }
+}

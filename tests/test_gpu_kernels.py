@@ -298,3 +298,41 @@ def test_apply_flat_dispatch_uses_hip():
         apply_mod.apply_flat("SGD", p_gpu, g_gpu, st_gpu, hyper)
         apply_mod.apply_dense("SGD", [p_cpu], [g_cpu], [st_cpu], hyper)
     assert torch.allclose(p_gpu.cpu(), p_cpu, atol=1e-6)
+
+
+def test_fused_adagrad(ext):
+    n = 100000
+    p = _rand(n, 2)
+    p_ref = torch.nn.Parameter(p.clone())
+    opt = torch.optim.Adagrad([p_ref], lr=5e-2, weight_decay=0.01,
+                              lr_decay=0.1)
+    acc = torch.zeros_like(p)
+    for step in range(1, 4):
+        g = _rand(n, 20 + step)
+        p_ref.grad = g.clone()
+        opt.step()
+        clr = 5e-2 / (1 + (step - 1) * 0.1)
+        ext.fused_adagrad(p, g, acc, clr, 1e-10, 0.01, False)
+    assert torch.allclose(p, p_ref.detach(), atol=1e-5), \
+        (p - p_ref.detach()).abs().max()
+
+
+@pytest.mark.parametrize("centered,momentum", [(False, 0.0), (True, 0.9),
+                                               (False, 0.9), (True, 0.0)])
+def test_fused_rmsprop(ext, centered, momentum):
+    n = 65536
+    p = _rand(n, 3)
+    p_ref = torch.nn.Parameter(p.clone())
+    opt = torch.optim.RMSprop([p_ref], lr=1e-3, alpha=0.95, momentum=momentum,
+                              centered=centered, weight_decay=0.02)
+    sq = torch.zeros_like(p)
+    ga = torch.zeros_like(p) if centered else None
+    buf = torch.zeros_like(p) if momentum else None
+    for step in range(3):
+        g = _rand(n, 30 + step)
+        p_ref.grad = g.clone()
+        opt.step()
+        ext.fused_rmsprop(p, g, sq, ga, buf, 1e-3, 0.95, 1e-8, 0.02,
+                          momentum, False)
+    assert torch.allclose(p, p_ref.detach(), atol=1e-5), \
+        (p - p_ref.detach()).abs().max()
