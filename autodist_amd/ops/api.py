@@ -63,6 +63,13 @@ def ext():
 
 # -- kernel wrappers (HIP on GPU, torch fallback on CPU) --------------------
 
+_FUSED_CLASSES = ("SGD", "Adam", "AdamW", "Adagrad", "RMSprop")
+
+
+def has_fused(cls_name: str) -> bool:
+    return cls_name in _FUSED_CLASSES
+
+
 def fused_apply(cls_name: str, param: torch.Tensor, grad: torch.Tensor,
                 state: dict, hyper: dict) -> bool:
     """One-kernel optimizer update over flat bucket buffers. Returns False

@@ -338,10 +338,29 @@ _APPLY = {"SGD": apply_sgd, "Adam": apply_adam, "AdamW": apply_adamw,
 
 
 def apply_dense(cls_name: str, params, grads, states, hyper):
-    """Apply one optimizer update to a list of params/shards (one fused
-    multi-tensor group)."""
+    """Apply one optimizer update to a list of params/shards. On GPU,
+    tensors route through the hand-written fused HIP kernels (one launch
+    per tensor instead of 3-5 _foreach passes); non-contiguous/odd-dtype
+    shards fall back to the torch multi-tensor path."""
     if not params:
         return
+    if params[0].is_cuda:
+        from autodist_amd.ops import api as ops_api
+        if ops_api.has_gpu_ops() and ops_api.has_fused(cls_name):
+            rest = ([], [], [])
+            for pp, gg, st in zip(params, grads, states):
+                if (pp.is_contiguous() and gg.is_contiguous()
+                        and pp.dtype == torch.float32
+                        and gg.dtype == torch.float32
+                        and ops_api.fused_apply(cls_name, pp, gg, st,
+                                                hyper)):
+                    continue
+                rest[0].append(pp)
+                rest[1].append(gg)
+                rest[2].append(st)
+            if rest[0]:
+                _APPLY[cls_name](rest[0], rest[1], rest[2], hyper)
+            return
     _APPLY[cls_name](params, grads, states, hyper)
 
 
