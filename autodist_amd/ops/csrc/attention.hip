@@ -61,31 +61,33 @@ __device__ __forceinline__ unsigned int drop_hash(unsigned int seed,
 // round-robin across the 8 XCDs, so all q-blocks of one (b,h) — which
 // re-read the SAME K/V — land on the SAME XCD's L2 when B*H % 8 == 0
 // (XCD-aware swizzle; B*H is the fast dimension mod 8).
-template <int RB>
+template <int RB, int D>
 __global__ void
-__launch_bounds__(256, 2)
+__launch_bounds__(256, D == 64 ? 2 : 1)
 attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                 const __hip_bfloat16* __restrict__ K,
                 const __hip_bfloat16* __restrict__ V,
                 __hip_bfloat16* __restrict__ O,
                 const float* __restrict__ mask, long S, long H, long NBH,
                 float scale, float p_drop, unsigned int seed) {
-  __shared__ short Ks[2][64][ATTN_D + KPAD];
-  __shared__ short VsT[2][ATTN_D][64 + KPAD];  // transposed: B-frag reads
-                                               // are one ds_read_b128
-  __shared__ short Pw[4][16][64 + KPAD];       // per-wave bf16 P staging
+  __shared__ short Ks[2][64][D + KPAD];
+  __shared__ short VsT[2][D][64 + KPAD];  // transposed: B-frag reads are
+                                          // one ds_read_b128
+  __shared__ short Pw[4][16][64 + KPAD];  // per-wave bf16 P staging
   int t = threadIdx.x;
   int w = t >> 6;        // wave 0..3
   int l = t & 63;        // lane
   long bh = (long)blockIdx.x % NBH;            // XCD swizzle (see above)
   long qb = (long)blockIdx.x / NBH;
   long q0 = qb * (64 * RB) + w * (16 * RB);    // this wave's RB*16 q rows
-  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * ATTN_D;
-  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * ATTN_D;
-  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * ATTN_D;
-  short* o_p = reinterpret_cast<short*>(O) + bh * S * ATTN_D;
+  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * D;
+  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * D;
+  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * D;
+  short* o_p = reinterpret_cast<short*>(O) + bh * S * D;
   const float* m_p = mask ? mask + (bh / H) * S : nullptr;
 
+  constexpr int NC = D / 32;   // A/B k-chunks along the head dim
+  constexpr int ND = D / 16;   // C d-tiles
   int am = l & 15;       // A-fragment m index / C column index
   int kg = l >> 4;       // k-group (0..3)
   const unsigned int thresh =
@@ -94,21 +96,21 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   const float scale2 = scale * ATTN_LOG2E;  // exp2-domain logit scale
 
   // Q fragments: q row (q0+rb*16+am) clamped for partial tiles
-  bf16x8_t qf[RB][2];
+  bf16x8_t qf[RB][NC];
 #pragma unroll
   for (int rb = 0; rb < RB; ++rb) {
     int qrow_a = q0 + rb * 16 + am < S ? (int)q0 + rb * 16 + am : (int)S - 1;
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
+    for (int c = 0; c < NC; ++c) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        qf[rb][c][j] = q_p[qrow_a * ATTN_D + c * 32 + kg * 8 + j];
+        qf[rb][c][j] = q_p[qrow_a * D + c * 32 + kg * 8 + j];
       }
     }
   }
 
   float m_run[RB][4], l_run[RB][4];
-  f32x4_t o_acc[RB][4];  // d-tiles of 16 cols each
+  f32x4_t o_acc[RB][ND];  // d-tiles of 16 cols each
 #pragma unroll
   for (int rb = 0; rb < RB; ++rb) {
 #pragma unroll
@@ -117,27 +119,28 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
       l_run[rb][r] = 0.f;
     }
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) o_acc[rb][dt] = {0.f, 0.f, 0.f, 0.f};
+    for (int dt = 0; dt < ND; ++dt) o_acc[rb][dt] = {0.f, 0.f, 0.f, 0.f};
   }
 
-  // cooperative 64x64 K/V stage: 256 threads x (one row of 64 shorts each
-  // as 2 bf16x8 loads) -> thread t stages row (t>>2) cols (t&3)*16..+15
-  int srow = t >> 2, scol = (t & 3) * 16;
+  // cooperative 64xD K/V stage: 256 threads x one row-quarter (D/4 shorts
+  // = D/32 bf16x8 loads) each -> thread t stages row (t>>2)
+  int srow = t >> 2, scol = (t & 3) * (D / 4);
+  constexpr int NH = D / 32;  // bf16x8 chunks per thread
 
-  bf16x8_t kreg[2], vreg[2];
+  bf16x8_t kreg[NH], vreg[NH];
   auto load_tile = [&](long kt) {  // issue global loads into registers
     int krow = kt + srow < S ? (int)kt + srow : (int)S - 1;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NH; ++half) {
       kreg[half] = *reinterpret_cast<const bf16x8_t*>(
-          &k_p[krow * ATTN_D + scol + half * 8]);
+          &k_p[krow * D + scol + half * 8]);
       vreg[half] = *reinterpret_cast<const bf16x8_t*>(
-          &v_p[krow * ATTN_D + scol + half * 8]);
+          &v_p[krow * D + scol + half * 8]);
     }
   };
   auto store_tile = [&](int buf) {  // registers -> LDS (after compute)
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NH; ++half) {
       *reinterpret_cast<bf16x8_t*>(&Ks[buf][srow][scol + half * 8]) =
           kreg[half];
 #pragma unroll
@@ -170,7 +173,7 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
       for (int h = 0; h < 4; ++h) {
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
+        for (int c = 0; c < NC; ++c) {
           bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
               &Ks[cur][h * 16 + am][c * 32 + kg * 8]);
           s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -212,7 +215,7 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
         }
       }
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
+      for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) o_acc[rb][dt][r] *= alpha[r];
       }
@@ -234,7 +237,7 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
       }
       // ---- O += P @ V : per 16-col d tile, two k-chunks of 32 keys
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
+      for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
         for (int kc = 0; kc < 2; ++kc) {
           bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
@@ -253,14 +256,14 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
   for (int rb = 0; rb < RB; ++rb) {
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+    for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long qrow = q0 + rb * 16 + kg * 4 + r;
         if (qrow >= S) continue;
         float val = o_acc[rb][dt][r] / l_run[rb][r] * rkeep;
         __hip_bfloat16 ob = __float2bfloat16(val);
-        o_p[qrow * ATTN_D + dt * 16 + am] = reinterpret_cast<short&>(ob);
+        o_p[qrow * D + dt * 16 + am] = reinterpret_cast<short&>(ob);
       }
     }
   }

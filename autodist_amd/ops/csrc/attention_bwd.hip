@@ -46,17 +46,19 @@ __device__ __forceinline__ unsigned int bwd_drop_hash(unsigned int seed,
   return x;
 }
 
-// load an A/B fragment row-block from LDS rows [64][64+BKPAD]
-__device__ __forceinline__ bwd_bf16x8 lds_frag(
-    const short (*buf)[ATTN_BD + BKPAD], int row, int col) {
+// load an A/B fragment row-block from LDS rows [64][D+BKPAD]
+template <int W>
+__device__ __forceinline__ bwd_bf16x8 lds_frag(const short (*buf)[W],
+                                               int row, int col) {
   return *reinterpret_cast<const bwd_bf16x8*>(&buf[row][col]);
 }
 
 // ---------------------------------------------------------------- K1
 // grid 1-D (ceil(S/64) * B*H), 256 threads (4 waves x 16 q rows).
 // Writes M, L (exp2-domain), delta [B*H, S] fp32 and dQ.
+template <int D>
 __global__ void
-__launch_bounds__(256, 2)
+__launch_bounds__(256, D == 64 ? 2 : 1)
 attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                   const __hip_bfloat16* __restrict__ K,
                   const __hip_bfloat16* __restrict__ V,
@@ -67,21 +69,23 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                   float* __restrict__ Dbuf,
                   const float* __restrict__ mask, long S, long H, long NBH,
                   float scale, float p_drop, unsigned int seed) {
-  __shared__ short Ks[64][ATTN_BD + BKPAD];
-  __shared__ short Vs[64][ATTN_BD + BKPAD];
-  __shared__ short KsT[ATTN_BD][64 + BKPAD];
+  __shared__ short Ks[64][D + BKPAD];
+  __shared__ short Vs[64][D + BKPAD];
+  __shared__ short KsT[D][64 + BKPAD];
   __shared__ short DSw[4][16][64 + BKPAD];  // per-wave bf16 dS staging
   int t = threadIdx.x;
   int w = t >> 6, l = t & 63;
   long bh = (long)blockIdx.x % NBH;
   long q0 = ((long)blockIdx.x / NBH) * 64 + w * 16;
-  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * ATTN_BD;
-  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * ATTN_BD;
-  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * ATTN_BD;
-  const short* o_p = reinterpret_cast<const short*>(O) + bh * S * ATTN_BD;
-  const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * ATTN_BD;
-  short* dq_p = reinterpret_cast<short*>(dQ) + bh * S * ATTN_BD;
+  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * D;
+  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * D;
+  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * D;
+  const short* o_p = reinterpret_cast<const short*>(O) + bh * S * D;
+  const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * D;
+  short* dq_p = reinterpret_cast<short*>(dQ) + bh * S * D;
   int am = l & 15, kg = l >> 4;
+  constexpr int NC = D / 32;
+  constexpr int ND = D / 16;
   const float* m_p = mask ? mask + (bh / H) * S : nullptr;
   const unsigned int thresh =
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
@@ -90,38 +94,38 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
   const float scale2 = scale * ATTN_LOG2E;  // exp2-domain logit scale
 
   int qrow_a = q0 + am < S ? (int)q0 + am : (int)S - 1;
-  bwd_bf16x8 qf[2], dof[2];
+  bwd_bf16x8 qf[NC], dof[NC];
 #pragma unroll
-  for (int c = 0; c < 2; ++c) {
+  for (int c = 0; c < NC; ++c) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      qf[c][j] = q_p[qrow_a * ATTN_BD + c * 32 + kg * 8 + j];
-      dof[c][j] = do_p[qrow_a * ATTN_BD + c * 32 + kg * 8 + j];
+      qf[c][j] = q_p[qrow_a * D + c * 32 + kg * 8 + j];
+      dof[c][j] = do_p[qrow_a * D + c * 32 + kg * 8 + j];
     }
   }
 
-  int srow = t >> 2, scol = (t & 3) * 16;
+  int srow = t >> 2, scol = (t & 3) * (D / 4);
   auto stage_k = [&](long kt) {
     int krow = kt + srow < S ? (int)kt + srow : (int)S - 1;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NC; ++half) {
       *reinterpret_cast<bwd_bf16x8*>(&Ks[srow][scol + half * 8]) =
           *reinterpret_cast<const bwd_bf16x8*>(
-              &k_p[krow * ATTN_BD + scol + half * 8]);
+              &k_p[krow * D + scol + half * 8]);
     }
   };
   auto stage_kvt = [&](long kt) {
     int krow = kt + srow < S ? (int)kt + srow : (int)S - 1;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NC; ++half) {
       bwd_bf16x8 kv = *reinterpret_cast<const bwd_bf16x8*>(
-          &k_p[krow * ATTN_BD + scol + half * 8]);
+          &k_p[krow * D + scol + half * 8]);
       *reinterpret_cast<bwd_bf16x8*>(&Ks[srow][scol + half * 8]) = kv;
 #pragma unroll
       for (int j = 0; j < 8; ++j) KsT[scol + half * 8 + j][srow] = kv[j];
       *reinterpret_cast<bwd_bf16x8*>(&Vs[srow][scol + half * 8]) =
           *reinterpret_cast<const bwd_bf16x8*>(
-              &v_p[krow * ATTN_BD + scol + half * 8]);
+              &v_p[krow * D + scol + half * 8]);
     }
   };
 
@@ -138,7 +142,7 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
     for (int h = 0; h < 4; ++h) {
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
+      for (int c = 0; c < NC; ++c) {
         bwd_bf16x8 kf = lds_frag(Ks, h * 16 + am, c * 32 + kg * 8);
         s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kf,
                                                            s_acc[h], 0, 0, 0);
@@ -175,12 +179,12 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
     long rowc = row < S ? row : S - 1;
     float part = 0.f;
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      int d = am * 4 + u;
+    for (int u = 0; u < D / 16; ++u) {  // D/16 cols per lane
+      int d = am * (D / 16) + u;
       float ov = __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
-          o_p)[rowc * ATTN_BD + d]);
+          o_p)[rowc * D + d]);
       float dv = __bfloat162float(reinterpret_cast<const __hip_bfloat16*>(
-          do_p)[rowc * ATTN_BD + d]);
+          do_p)[rowc * D + d]);
       part += ov * dv;
     }
     delta[r] = dpp16_sum(part);
@@ -191,9 +195,9 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
     }
   }
   // ---- pass B: dQ accumulation
-  bwd_f32x4 dq_acc[4];
+  bwd_f32x4 dq_acc[ND];
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt) dq_acc[dt] = {0, 0, 0, 0};
+  for (int dt = 0; dt < ND; ++dt) dq_acc[dt] = {0, 0, 0, 0};
   for (long kt = 0; kt < S; kt += 64) {
     __syncthreads();
     stage_kvt(kt);
@@ -206,7 +210,7 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
     for (int h = 0; h < 4; ++h) {
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
+      for (int c = 0; c < NC; ++c) {
         bwd_bf16x8 kf = lds_frag(Ks, h * 16 + am, c * 32 + kg * 8);
         bwd_bf16x8 vf = lds_frag(Vs, h * 16 + am, c * 32 + kg * 8);
         s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[c], kf,
@@ -245,7 +249,7 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
           &DSw[w][am][kc * 32 + kg * 8]);
     // dQ += dS K : m=q, n=d(16/tile), k=keys(64 in 2 chunks)
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+    for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         bwd_bf16x8 kfT = *reinterpret_cast<const bwd_bf16x8*>(
@@ -257,13 +261,13 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
     wave_lds_fence();
   }
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt) {
+  for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       long row = q0 + kg * 4 + r;
       if (row >= S) continue;
       __hip_bfloat16 b = __float2bfloat16(dq_acc[dt][r]);
-      dq_p[row * ATTN_BD + dt * 16 + am] = reinterpret_cast<short&>(b);
+      dq_p[row * D + dt * 16 + am] = reinterpret_cast<short&>(b);
     }
   }
 }
@@ -271,8 +275,9 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
 // ---------------------------------------------------------------- K2
 // grid 1-D (ceil(S/64) * B*H), 256 threads (4 waves x 16 keys): per
 // 64-KEY block, accumulate dK and dV over all q (Q/dO tiles via LDS).
+template <int D>
 __global__ void
-__launch_bounds__(256, 2)
+__launch_bounds__(256, D == 64 ? 2 : 1)
 attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
                    const __hip_bfloat16* __restrict__ K,
                    const __hip_bfloat16* __restrict__ V,
@@ -284,23 +289,25 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
                    const float* __restrict__ Dbuf,
                    const float* __restrict__ mask, long S, long H, long NBH,
                    float scale, float p_drop, unsigned int seed) {
-  __shared__ short Qs[64][ATTN_BD + BKPAD];
-  __shared__ short dOs[64][ATTN_BD + BKPAD];
-  __shared__ short QsT[ATTN_BD][64 + BKPAD];
-  __shared__ short dOsT[ATTN_BD][64 + BKPAD];
+  __shared__ short Qs[64][D + BKPAD];
+  __shared__ short dOs[64][D + BKPAD];
+  __shared__ short QsT[D][64 + BKPAD];
+  __shared__ short dOsT[D][64 + BKPAD];
   __shared__ short Pw[4][16][64 + BKPAD];   // P' staging (bf16)
   __shared__ short DSw[4][16][64 + BKPAD];  // dS' staging (bf16)
   int t = threadIdx.x;
   int w = t >> 6, l = t & 63;
   long bh = (long)blockIdx.x % NBH;
   long k0 = ((long)blockIdx.x / NBH) * 64 + w * 16;  // this wave's 16 keys
-  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * ATTN_BD;
-  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * ATTN_BD;
-  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * ATTN_BD;
-  const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * ATTN_BD;
-  short* dk_p = reinterpret_cast<short*>(dK) + bh * S * ATTN_BD;
-  short* dv_p = reinterpret_cast<short*>(dV) + bh * S * ATTN_BD;
+  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * D;
+  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * D;
+  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * D;
+  const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * D;
+  short* dk_p = reinterpret_cast<short*>(dK) + bh * S * D;
+  short* dv_p = reinterpret_cast<short*>(dV) + bh * S * D;
   int am = l & 15, kg = l >> 4;
+  constexpr int NC = D / 32;
+  constexpr int ND = D / 16;
   const float* m_p = mask ? mask + (bh / H) * S : nullptr;
   const unsigned int thresh =
       (unsigned int)fminf(p_drop * 4294967296.0f, 4294967040.0f);
@@ -309,13 +316,13 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
   const float scale2 = scale * ATTN_LOG2E;
 
   int krow_a = k0 + am < S ? (int)k0 + am : (int)S - 1;
-  bwd_bf16x8 kf[2], vf[2];
+  bwd_bf16x8 kf[NC], vf[NC];
 #pragma unroll
-  for (int c = 0; c < 2; ++c) {
+  for (int c = 0; c < NC; ++c) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      kf[c][j] = k_p[krow_a * ATTN_BD + c * 32 + kg * 8 + j];
-      vf[c][j] = v_p[krow_a * ATTN_BD + c * 32 + kg * 8 + j];
+      kf[c][j] = k_p[krow_a * D + c * 32 + kg * 8 + j];
+      vf[c][j] = v_p[krow_a * D + c * 32 + kg * 8 + j];
     }
   }
   // additive mask value of this lane's OWN key rows (kg*4+r)
@@ -326,25 +333,25 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
     mvk2[r] = (m_p && key < S) ? m_p[key] * ATTN_LOG2E : 0.f;
   }
 
-  bwd_f32x4 dk_acc[4], dv_acc[4];
+  bwd_f32x4 dk_acc[ND], dv_acc[ND];
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt) {
+  for (int dt = 0; dt < ND; ++dt) {
     dk_acc[dt] = {0, 0, 0, 0};
     dv_acc[dt] = {0, 0, 0, 0};
   }
 
-  int srow = t >> 2, scol = (t & 3) * 16;
+  int srow = t >> 2, scol = (t & 3) * (D / 4);
   auto stage_qdo = [&](long qt) {
     int qrow = qt + srow < S ? (int)qt + srow : (int)S - 1;
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
+    for (int half = 0; half < NC; ++half) {
       bwd_bf16x8 qv = *reinterpret_cast<const bwd_bf16x8*>(
-          &q_p[qrow * ATTN_BD + scol + half * 8]);
+          &q_p[qrow * D + scol + half * 8]);
       *reinterpret_cast<bwd_bf16x8*>(&Qs[srow][scol + half * 8]) = qv;
 #pragma unroll
       for (int j = 0; j < 8; ++j) QsT[scol + half * 8 + j][srow] = qv[j];
       bwd_bf16x8 dv8 = *reinterpret_cast<const bwd_bf16x8*>(
-          &do_p[qrow * ATTN_BD + scol + half * 8]);
+          &do_p[qrow * D + scol + half * 8]);
       *reinterpret_cast<bwd_bf16x8*>(&dOs[srow][scol + half * 8]) = dv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) dOsT[scol + half * 8 + j][srow] = dv8[j];
@@ -363,7 +370,7 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
     for (int h = 0; h < 4; ++h) {
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
+      for (int c = 0; c < NC; ++c) {
         bwd_bf16x8 qfr = lds_frag(Qs, h * 16 + am, c * 32 + kg * 8);
         bwd_bf16x8 dofr = lds_frag(dOs, h * 16 + am, c * 32 + kg * 8);
         s_acc[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[c], qfr,
@@ -410,7 +417,7 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
     }
     // dV += P' dO : m=key, n=d, k=q ; dK += dS' Q : m=key, n=d, k=q
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+    for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
       for (int kc = 0; kc < 2; ++kc) {
         bwd_bf16x8 dofT = *reinterpret_cast<const bwd_bf16x8*>(
@@ -426,15 +433,15 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
     wave_lds_fence();
   }
 #pragma unroll
-  for (int dt = 0; dt < 4; ++dt) {
+  for (int dt = 0; dt < ND; ++dt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       long row = k0 + kg * 4 + r;
       if (row >= S) continue;
       __hip_bfloat16 bk = __float2bfloat16(dk_acc[dt][r]);
       __hip_bfloat16 bv = __float2bfloat16(dv_acc[dt][r]);
-      dk_p[row * ATTN_BD + dt * 16 + am] = reinterpret_cast<short&>(bk);
-      dv_p[row * ATTN_BD + dt * 16 + am] = reinterpret_cast<short&>(bv);
+      dk_p[row * D + dt * 16 + am] = reinterpret_cast<short&>(bk);
+      dv_p[row * D + dt * 16 + am] = reinterpret_cast<short&>(bv);
     }
   }
 }

@@ -466,8 +466,8 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        double scale,
                        c10::optional<torch::Tensor> mask = c10::nullopt,
                        double p_drop = 0.0, int64_t seed = 0) {
-  TORCH_CHECK(q.dim() == 4 && q.size(3) == ATTN_D,
-              "attn_fwd expects [B,H,S,64]");
+  TORCH_CHECK(q.dim() == 4 && (q.size(3) == 64 || q.size(3) == 128),
+              "attn_fwd expects [B,H,S,64|128]");
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   long B = q.size(0), H = q.size(1), S = q.size(2);
@@ -478,26 +478,22 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   // RB=1 keeps more blocks in flight for short ones. 1-D grid with bh as
   // the fast dimension = XCD-locality swizzle (see kernel comment).
   long nbh = B * H;
-  if (S >= 256) {
-    long nqb = (S + 127) / 128;
-    hipLaunchKernelGGL(attn_fwd_kernel<2>, dim3(nqb * nbh), dim3(256), 0,
-                       cur_stream(),
+  auto launch = [&](auto kern, long rows) {
+    long nqb = (S + rows - 1) / rows;
+    hipLaunchKernelGGL(kern, dim3(nqb * nbh), dim3(256), 0, cur_stream(),
                        reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
                        attn_mask_ptr(mask, B, S), S, H, nbh, (float)scale,
                        (float)p_drop, (unsigned int)(uint64_t)seed);
+  };
+  if (q.size(3) == 64) {
+    if (S >= 256) launch(attn_fwd_kernel<2, 64>, 128);
+    else launch(attn_fwd_kernel<1, 64>, 64);
   } else {
-    long nqb = (S + 63) / 64;
-    hipLaunchKernelGGL(attn_fwd_kernel<1>, dim3(nqb * nbh), dim3(256), 0,
-                       cur_stream(),
-                       reinterpret_cast<__hip_bfloat16*>(q.data_ptr()),
-                       reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
-                       reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
-                       reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
-                       attn_mask_ptr(mask, B, S), S, H, nbh, (float)scale,
-                       (float)p_drop, (unsigned int)(uint64_t)seed);
+    if (S >= 256) launch(attn_fwd_kernel<2, 128>, 128);
+    else launch(attn_fwd_kernel<1, 128>, 64);
   }
   return o;
 }
@@ -518,7 +514,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     c10::optional<torch::Tensor> mask
                                         = c10::nullopt,
                                     double p_drop = 0.0, int64_t seed = 0) {
-  TORCH_CHECK(q.dim() == 4 && q.size(3) == ATTN_BD);
+  TORCH_CHECK(q.dim() == 4 && (q.size(3) == 64 || q.size(3) == 128));
   TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
   long B = q.size(0), H = q.size(1), S = q.size(2);
   TORCH_CHECK(S % 32 == 0 && S >= 32);
@@ -536,16 +532,29 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   dim3 grid(((S + 63) / 64) * nbh);
   auto st = cur_stream();
 #define BF16P(t) reinterpret_cast<__hip_bfloat16*>((t).data_ptr())
-  hipLaunchKernelGGL(attn_bwd_q_kernel, grid, dim3(256), 0, st, BF16P(q),
-                     BF16P(k), BF16P(v), BF16P(o), BF16P(dout), BF16P(dq),
-                     Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
-                     Dbuf.data_ptr<float>(), mp, S, H, nbh, (float)scale,
-                     (float)p_drop, sd);
-  hipLaunchKernelGGL(attn_bwd_kv_kernel, grid, dim3(256), 0, st, BF16P(q),
-                     BF16P(k), BF16P(v), BF16P(dout), BF16P(dk), BF16P(dv),
-                     Mbuf.data_ptr<float>(), Lbuf.data_ptr<float>(),
-                     Dbuf.data_ptr<float>(), mp, S, H, nbh, (float)scale,
-                     (float)p_drop, sd);
+  if (q.size(3) == 64) {
+    hipLaunchKernelGGL(attn_bwd_q_kernel<64>, grid, dim3(256), 0, st,
+                       BF16P(q), BF16P(k), BF16P(v), BF16P(o), BF16P(dout),
+                       BF16P(dq), Mbuf.data_ptr<float>(),
+                       Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
+                       S, H, nbh, (float)scale, (float)p_drop, sd);
+    hipLaunchKernelGGL(attn_bwd_kv_kernel<64>, grid, dim3(256), 0, st,
+                       BF16P(q), BF16P(k), BF16P(v), BF16P(dout), BF16P(dk),
+                       BF16P(dv), Mbuf.data_ptr<float>(),
+                       Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
+                       S, H, nbh, (float)scale, (float)p_drop, sd);
+  } else {
+    hipLaunchKernelGGL(attn_bwd_q_kernel<128>, grid, dim3(256), 0, st,
+                       BF16P(q), BF16P(k), BF16P(v), BF16P(o), BF16P(dout),
+                       BF16P(dq), Mbuf.data_ptr<float>(),
+                       Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
+                       S, H, nbh, (float)scale, (float)p_drop, sd);
+    hipLaunchKernelGGL(attn_bwd_kv_kernel<128>, grid, dim3(256), 0, st,
+                       BF16P(q), BF16P(k), BF16P(v), BF16P(dout), BF16P(dk),
+                       BF16P(dv), Mbuf.data_ptr<float>(),
+                       Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
+                       S, H, nbh, (float)scale, (float)p_drop, sd);
+  }
 #undef BF16P
   return {dq, dk, dv};
 }

@@ -41,7 +41,8 @@ def can_use_fused(q, attn_mask, dropout_p) -> bool:
     """Kernel applicability (training AND inference; reference context: the
     reference leaves attention to TF ops — hand-written CDNA4 here is the
     north-star requirement)."""
-    if not (q.is_cuda and q.dtype == torch.bfloat16 and q.size(-1) == 64
+    if not (q.is_cuda and q.dtype == torch.bfloat16
+            and q.size(-1) in (64, 128)
             and q.size(-2) % 32 == 0 and q.size(-2) >= 32
             and 0.0 <= dropout_p < 1.0 and ops_api.has_gpu_ops()):
         return False
