@@ -323,6 +323,17 @@ def main():
 
     if rank == 0:
         value = world * wl["items_per_step"] * args.steps / dt
+        # MIOpen find-cache sanity: with a cold find
+        # cache and too little warmup, immediate-mode picks naive wrw conv
+        # kernels and the metric silently collapses ~40x (measured 214 vs
+        # 8458 img/s). Flag the anomaly instead of reporting it silently.
+        if use_cuda and args.model.startswith("resnet"):
+            per_img_ms = dt / args.steps * 1e3 / wl["items_per_step"]
+            if per_img_ms > 0.6:  # healthy: ~0.12 ms/img (b512, MI355X)
+                print(f"# WARNING: {per_img_ms:.2f} ms/image is ~5x+ off the "
+                      f"expected MI355X rate — MIOpen likely missed its "
+                      f"exhaustive find (cold cache / too few warmup steps);"
+                      f" rerun with --warmup >= 5", file=sys.stderr)
         result = {
             "metric": wl["metric"],
             "value": round(value, 2),
