@@ -66,17 +66,19 @@ class BertSelfAttention(nn.Module):
 class BertLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
+        from autodist_amd.ops.fused_ln import FusedLayerNorm
         self.attn = BertSelfAttention(cfg)
-        self.ln1 = nn.LayerNorm(cfg.hidden, eps=1e-12)
+        self.ln1 = FusedLayerNorm(cfg.hidden, eps=1e-12)
         self.fc1 = nn.Linear(cfg.hidden, cfg.intermediate)
         self.fc2 = nn.Linear(cfg.intermediate, cfg.hidden)
-        self.ln2 = nn.LayerNorm(cfg.hidden, eps=1e-12)
+        self.ln2 = FusedLayerNorm(cfg.hidden, eps=1e-12)
         self.drop = nn.Dropout(cfg.dropout)
 
     def forward(self, x, attn_mask=None):
-        x = self.ln1(x + self.drop(self.attn(x, attn_mask)))
+        # residual adds fold into the fused bf16 LN kernels (gfx950)
+        x = self.ln1(self.drop(self.attn(x, attn_mask)), residual=x)
         h = self.fc2(torch.nn.functional.gelu(self.fc1(x)))
-        return self.ln2(x + self.drop(h))
+        return self.ln2(self.drop(h), residual=x)
 
 
 class BertModel(nn.Module):
@@ -87,7 +89,8 @@ class BertModel(nn.Module):
                                     sparse=sparse_embeddings)
         self.pos_emb = nn.Embedding(cfg.max_seq, cfg.hidden)
         self.seg_emb = nn.Embedding(cfg.type_vocab, cfg.hidden)
-        self.emb_ln = nn.LayerNorm(cfg.hidden, eps=1e-12)
+        from autodist_amd.ops.fused_ln import FusedLayerNorm
+        self.emb_ln = FusedLayerNorm(cfg.hidden, eps=1e-12)
         self.emb_drop = nn.Dropout(cfg.dropout)
         self.layers = nn.ModuleList(BertLayer(cfg) for _ in range(cfg.layers))
         self.apply(self._init)
@@ -120,8 +123,9 @@ class BertForPreTraining(nn.Module):
     def __init__(self, cfg: BertConfig, sparse_embeddings=False):
         super().__init__()
         self.bert = BertModel(cfg, sparse_embeddings)
+        from autodist_amd.ops.fused_ln import FusedLayerNorm
         self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
-        self.mlm_ln = nn.LayerNorm(cfg.hidden, eps=1e-12)
+        self.mlm_ln = FusedLayerNorm(cfg.hidden, eps=1e-12)
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
         self.nsp = nn.Linear(cfg.hidden, 2)
 

@@ -10,6 +10,7 @@
 #include "mfma_probe.hip"
 #include "attention.hip"
 #include "attention_bwd.hip"
+#include "ln_ops.hip"
 
 namespace {
 
@@ -559,6 +560,63 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   return {dq, dk, dv};
 }
 
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x,
+                                  c10::optional<torch::Tensor> res,
+                                  torch::Tensor gamma, torch::Tensor beta,
+                                  double eps) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  long H = x.size(-1);
+  long N = x.numel() / H;
+  TORCH_CHECK(H <= 4096, "ln_fwd supports H <= 4096");
+  TORCH_CHECK(gamma.scalar_type() == torch::kFloat32 && gamma.is_contiguous());
+  auto y = torch::empty_like(x);
+  bool has_res = res.has_value() && res->defined();
+  torch::Tensor u = has_res ? torch::empty_like(x) : x;
+  auto fopt = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({N}, fopt);
+  auto rstd = torch::empty({N}, fopt);
+  const __hip_bfloat16* rp = has_res
+      ? reinterpret_cast<__hip_bfloat16*>(res->data_ptr()) : nullptr;
+  __hip_bfloat16* up = has_res
+      ? reinterpret_cast<__hip_bfloat16*>(u.data_ptr()) : nullptr;
+  hipLaunchKernelGGL(ln_fwd_kernel, dim3(N), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(x.data_ptr()), rp,
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(y.data_ptr()), up,
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), N,
+                     (int)H, (float)eps);
+  return {y, u, mean, rstd};
+}
+
+std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor u,
+                                  torch::Tensor gamma, torch::Tensor mean,
+                                  torch::Tensor rstd) {
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 && dy.is_contiguous());
+  long H = dy.size(-1);
+  long N = dy.numel() / H;
+  auto dx = torch::empty_like(dy);
+  hipLaunchKernelGGL(ln_bwd_dx_kernel, dim3(N), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(dy.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(u.data_ptr()),
+                     gamma.data_ptr<float>(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()), N,
+                     (int)H);
+  // chunked dgamma/dbeta partials, reduced on the torch side
+  long rows_per_chunk = 256;
+  long nchunks = (N + rows_per_chunk - 1) / rows_per_chunk;
+  auto partials = torch::empty({nchunks, 2, H},
+                               dy.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(ln_bwd_gb_kernel, dim3((H + 63) / 64, nchunks),
+                     dim3(256), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(dy.data_ptr()),
+                     reinterpret_cast<__hip_bfloat16*>(u.data_ptr()),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     partials.data_ptr<float>(), N, (int)H, rows_per_chunk);
+  auto gb = partials.sum(0);
+  return {dx, gb[0], gb[1]};
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B, long cand) {
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32);
   TORCH_CHECK(B.scalar_type() == torch::kBFloat16 && B.numel() == 32 * 16);
@@ -609,6 +667,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "padded M = flat + err (zero tail)");
   m.def("mfma_probe", &mfma_probe,
         "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
+  m.def("ln_fwd", &ln_fwd,
+        "fused bf16 LayerNorm(+residual) forward -> y, u, mean, rstd");
+  m.def("ln_bwd", &ln_bwd, "fused bf16 LayerNorm backward -> dx, dgamma, "
+        "dbeta");
   m.def("attn_dropmask", &attn_dropmask,
         "materialize the hash dropout keep-mask (testing)");
   m.def("attn_fwd", &attn_fwd,

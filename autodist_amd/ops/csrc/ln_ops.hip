@@ -1,0 +1,168 @@
+// Fused LayerNorm (+ residual add) for bf16 transformer streams, gfx950.
+//
+// torch's LN on a bf16 autocast stream runs fp32 kernels bracketed by
+// bfloat16<->float32 copy kernels (measured: the cast/elementwise cluster
+// around LN was ~13% of the BERT-base step). These kernels read/write
+// bf16 directly with fp32 accumulation, and fold the residual add
+// y = LN(a + b) into the same pass (saving the separate add kernel AND
+// its extra HBM round trip). Saved for backward: the summed input u
+// (bf16, doubles as the residual stream), mean and rstd (fp32 per row).
+//
+// Shapes: x [N, H] rows normalized over H; H <= 256 * LN_MAX_PER_THREAD
+// and H % 2 == 0 (bf16 pairs). One 256-thread block per row; row sums
+// via DPP 16-lane reduction + 2 cross-group shuffles + LDS across the 4
+// waves.
+#include "common.h"
+
+#define LN_MAX_PER_THREAD 16  // H up to 4096
+
+typedef short ln_bf16x2 __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ float ln_wave_sum(float v) {
+  v = dpp16_sum(v);
+  v += __shfl_xor(v, 16, 64);
+  v += __shfl_xor(v, 32, 64);
+  return v;
+}
+
+// block-wide sum of two values at once (saves one barrier round)
+__device__ __forceinline__ void ln_block_sum2(float& a, float& b,
+                                              float* lds /* >= 8 */) {
+  int w = threadIdx.x >> 6;
+  a = ln_wave_sum(a);
+  b = ln_wave_sum(b);
+  if ((threadIdx.x & 63) == 0) {
+    lds[w] = a;
+    lds[4 + w] = b;
+  }
+  __syncthreads();
+  a = lds[0] + lds[1] + lds[2] + lds[3];
+  b = lds[4] + lds[5] + lds[6] + lds[7];
+  __syncthreads();
+}
+
+__global__ void
+__launch_bounds__(256)
+ln_fwd_kernel(const __hip_bfloat16* __restrict__ x,
+              const __hip_bfloat16* __restrict__ res,  // nullable
+              const float* __restrict__ gamma, const float* __restrict__ beta,
+              __hip_bfloat16* __restrict__ y,
+              __hip_bfloat16* __restrict__ u_out,      // nullable
+              float* __restrict__ mean_out, float* __restrict__ rstd_out,
+              long N, int H, float eps) {
+  __shared__ float lds[8];
+  long row = blockIdx.x;
+  if (row >= N) return;
+  const __hip_bfloat16* xr = x + row * H;
+  const __hip_bfloat16* rr = res ? res + row * H : nullptr;
+  float v[LN_MAX_PER_THREAD];
+  int nper = (H + 255) >> 8;
+  float acc = 0.f, acc2 = 0.f;
+  for (int k = 0; k < nper; ++k) {
+    int i = threadIdx.x + (k << 8);
+    float u = 0.f;
+    if (i < H) {
+      u = __bfloat162float(xr[i]);
+      if (rr) u += __bfloat162float(rr[i]);
+    }
+    v[k] = u;
+    acc += u;
+    acc2 += u * u;
+  }
+  ln_block_sum2(acc, acc2, lds);
+  float mean = acc / H;
+  float var = fmaxf(acc2 / H - mean * mean, 0.f);
+  float rstd = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  __hip_bfloat16* yr = y + row * H;
+  __hip_bfloat16* ur = u_out ? u_out + row * H : nullptr;
+  for (int k = 0; k < nper; ++k) {
+    int i = threadIdx.x + (k << 8);
+    if (i < H) {
+      float o = (v[k] - mean) * rstd * gamma[i] + beta[i];
+      yr[i] = __float2bfloat16(o);
+      if (ur) ur[i] = __float2bfloat16(v[k]);
+    }
+  }
+}
+
+// dx = rstd * (g - mean(g) - xhat * mean(g*xhat)), g = dy*gamma (fp32)
+__global__ void
+__launch_bounds__(256)
+ln_bwd_dx_kernel(const __hip_bfloat16* __restrict__ dy,
+                 const __hip_bfloat16* __restrict__ u,
+                 const float* __restrict__ gamma,
+                 const float* __restrict__ mean_in,
+                 const float* __restrict__ rstd_in,
+                 __hip_bfloat16* __restrict__ dx, long N, int H) {
+  __shared__ float lds[8];
+  long row = blockIdx.x;
+  if (row >= N) return;
+  const __hip_bfloat16* dyr = dy + row * H;
+  const __hip_bfloat16* ur = u + row * H;
+  float mean = mean_in[row], rstd = rstd_in[row];
+  float g[LN_MAX_PER_THREAD], xh[LN_MAX_PER_THREAD];
+  int nper = (H + 255) >> 8;
+  float c1 = 0.f, c2 = 0.f;
+  for (int k = 0; k < nper; ++k) {
+    int i = threadIdx.x + (k << 8);
+    float gv = 0.f, xv = 0.f;
+    if (i < H) {
+      gv = __bfloat162float(dyr[i]) * gamma[i];
+      xv = (__bfloat162float(ur[i]) - mean) * rstd;
+    }
+    g[k] = gv;
+    xh[k] = xv;
+    c1 += gv;
+    c2 += gv * xv;
+  }
+  ln_block_sum2(c1, c2, lds);
+  c1 /= H;
+  c2 /= H;
+  __hip_bfloat16* dxr = dx + row * H;
+  for (int k = 0; k < nper; ++k) {
+    int i = threadIdx.x + (k << 8);
+    if (i < H)
+      dxr[i] = __float2bfloat16(rstd * (g[k] - c1 - xh[k] * c2));
+  }
+}
+
+// per-chunk partial dgamma/dbeta: grid (H/64, nchunks); 256 threads =
+// 64 cols x 4 row-lanes; partials [nchunks, 2, H] fp32 summed by caller.
+__global__ void
+__launch_bounds__(256)
+ln_bwd_gb_kernel(const __hip_bfloat16* __restrict__ dy,
+                 const __hip_bfloat16* __restrict__ u,
+                 const float* __restrict__ mean_in,
+                 const float* __restrict__ rstd_in,
+                 float* __restrict__ partials, long N, int H,
+                 long rows_per_chunk) {
+  __shared__ float pg[4][64];
+  __shared__ float pb[4][64];
+  int col = (blockIdx.x << 6) + (threadIdx.x & 63);
+  int rl = threadIdx.x >> 6;  // row-lane 0..3
+  long r0 = (long)blockIdx.y * rows_per_chunk;
+  long r1 = min(r0 + rows_per_chunk, N);
+  float sg = 0.f, sb = 0.f;
+  if (col < H) {
+    for (long r = r0 + rl; r < r1; r += 4) {
+      float d = __bfloat162float(dy[r * H + col]);
+      float xh = (__bfloat162float(u[r * H + col]) - mean_in[r]) * rstd_in[r];
+      sg += d * xh;
+      sb += d;
+    }
+  }
+  pg[rl][threadIdx.x & 63] = sg;
+  pb[rl][threadIdx.x & 63] = sb;
+  __syncthreads();
+  if (rl == 0 && col < H) {
+    int c = threadIdx.x & 63;
+    float tg = pg[0][c] + pg[1][c] + pg[2][c] + pg[3][c];
+    float tb = pb[0][c] + pb[1][c] + pb[2][c] + pb[3][c];
+    partials[((long)blockIdx.y * 2) * H + col] = tg;
+    partials[((long)blockIdx.y * 2 + 1) * H + col] = tb;
+  }
+}
