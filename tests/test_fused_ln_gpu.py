@@ -30,7 +30,9 @@ def test_ln_fwd_bwd_vs_torch(ext, N, H):
     yr = torch.nn.functional.layer_norm(xf, (H,), gf, bf, 1e-12)
     yr.backward(dy.float())
     assert (y.float() - yr.detach()).abs().max().item() < 3e-2
-    assert torch.allclose(u.float(), x.float() + res.float(), atol=1e-2)
+    # u is stored bf16: allow bf16 quantization of the fp32 sum
+    assert torch.allclose(u.float(), x.float() + res.float(), atol=1e-2,
+                          rtol=1e-2)
     dx, dgamma, dbeta = ext.ln_bwd(dy, u, gamma, mean, rstd)
     assert (dx.float() - xf.grad).abs().max().item() < 3e-2, \
         (dx.float() - xf.grad).abs().max()
