@@ -69,6 +69,9 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
                 const __hip_bfloat16* __restrict__ V,
                 __hip_bfloat16* __restrict__ O,
                 const float* __restrict__ mask, long S, long H, long NBH,
+                long q_sb, long q_sh, long q_ss,  // element strides: Q/K/V
+                long k_sb, long k_sh, long k_ss,  // may be VIEWS of the
+                long v_sb, long v_sh, long v_ss,  // fused qkv projection
                 float scale, float p_drop, unsigned int seed) {
   __shared__ short Ks[2][64][D + KPAD];
   __shared__ short VsT[2][D][64 + KPAD];  // transposed: B-frag reads are
@@ -80,9 +83,10 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
   long bh = (long)blockIdx.x % NBH;            // XCD swizzle (see above)
   long qb = (long)blockIdx.x / NBH;
   long q0 = qb * (64 * RB) + w * (16 * RB);    // this wave's RB*16 q rows
-  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * D;
-  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * D;
-  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * D;
+  long b = bh / H, hh = bh % H;
+  const short* q_p = reinterpret_cast<const short*>(Q) + b * q_sb + hh * q_sh;
+  const short* k_p = reinterpret_cast<const short*>(K) + b * k_sb + hh * k_sh;
+  const short* v_p = reinterpret_cast<const short*>(V) + b * v_sb + hh * v_sh;
   short* o_p = reinterpret_cast<short*>(O) + bh * S * D;
   const float* m_p = mask ? mask + (bh / H) * S : nullptr;
 
@@ -104,7 +108,7 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
     for (int c = 0; c < NC; ++c) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        qf[rb][c][j] = q_p[qrow_a * D + c * 32 + kg * 8 + j];
+        qf[rb][c][j] = q_p[qrow_a * q_ss + c * 32 + kg * 8 + j];
       }
     }
   }
@@ -133,9 +137,9 @@ attn_fwd_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
     for (int half = 0; half < NH; ++half) {
       kreg[half] = *reinterpret_cast<const bf16x8_t*>(
-          &k_p[krow * D + scol + half * 8]);
+          &k_p[krow * k_ss + scol + half * 8]);
       vreg[half] = *reinterpret_cast<const bf16x8_t*>(
-          &v_p[krow * D + scol + half * 8]);
+          &v_p[krow * v_ss + scol + half * 8]);
     }
   };
   auto store_tile = [&](int buf) {  // registers -> LDS (after compute)

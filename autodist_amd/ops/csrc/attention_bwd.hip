@@ -68,6 +68,8 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
                   float* __restrict__ Mbuf, float* __restrict__ Lbuf,
                   float* __restrict__ Dbuf,
                   const float* __restrict__ mask, long S, long H, long NBH,
+                  long q_sb, long q_sh, long q_ss, long k_sb, long k_sh,
+                  long k_ss, long v_sb, long v_sh, long v_ss,
                   float scale, float p_drop, unsigned int seed) {
   __shared__ short Ks[64][D + BKPAD];
   __shared__ short Vs[64][D + BKPAD];
@@ -77,9 +79,13 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
   int w = t >> 6, l = t & 63;
   long bh = (long)blockIdx.x % NBH;
   long q0 = ((long)blockIdx.x / NBH) * 64 + w * 16;
-  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * D;
-  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * D;
-  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * D;
+  long bb = bh / H, hh = bh % H;
+  const short* q_p = reinterpret_cast<const short*>(Q) + bb * q_sb
+                     + hh * q_sh;
+  const short* k_p = reinterpret_cast<const short*>(K) + bb * k_sb
+                     + hh * k_sh;
+  const short* v_p = reinterpret_cast<const short*>(V) + bb * v_sb
+                     + hh * v_sh;
   const short* o_p = reinterpret_cast<const short*>(O) + bh * S * D;
   const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * D;
   short* dq_p = reinterpret_cast<short*>(dQ) + bh * S * D;
@@ -99,7 +105,7 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
   for (int c = 0; c < NC; ++c) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      qf[c][j] = q_p[qrow_a * D + c * 32 + kg * 8 + j];
+      qf[c][j] = q_p[qrow_a * q_ss + c * 32 + kg * 8 + j];
       dof[c][j] = do_p[qrow_a * D + c * 32 + kg * 8 + j];
     }
   }
@@ -111,7 +117,7 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
     for (int half = 0; half < NC; ++half) {
       *reinterpret_cast<bwd_bf16x8*>(&Ks[srow][scol + half * 8]) =
           *reinterpret_cast<const bwd_bf16x8*>(
-              &k_p[krow * D + scol + half * 8]);
+              &k_p[krow * k_ss + scol + half * 8]);
     }
   };
   auto stage_kvt = [&](long kt) {
@@ -119,13 +125,13 @@ attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
     for (int half = 0; half < NC; ++half) {
       bwd_bf16x8 kv = *reinterpret_cast<const bwd_bf16x8*>(
-          &k_p[krow * D + scol + half * 8]);
+          &k_p[krow * k_ss + scol + half * 8]);
       *reinterpret_cast<bwd_bf16x8*>(&Ks[srow][scol + half * 8]) = kv;
 #pragma unroll
       for (int j = 0; j < 8; ++j) KsT[scol + half * 8 + j][srow] = kv[j];
       *reinterpret_cast<bwd_bf16x8*>(&Vs[srow][scol + half * 8]) =
           *reinterpret_cast<const bwd_bf16x8*>(
-              &v_p[krow * D + scol + half * 8]);
+              &v_p[krow * v_ss + scol + half * 8]);
     }
   };
 
@@ -288,6 +294,8 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
                    const float* __restrict__ Lbuf,
                    const float* __restrict__ Dbuf,
                    const float* __restrict__ mask, long S, long H, long NBH,
+                   long q_sb, long q_sh, long q_ss, long k_sb, long k_sh,
+                   long k_ss, long v_sb, long v_sh, long v_ss,
                    float scale, float p_drop, unsigned int seed) {
   __shared__ short Qs[64][D + BKPAD];
   __shared__ short dOs[64][D + BKPAD];
@@ -299,9 +307,13 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
   int w = t >> 6, l = t & 63;
   long bh = (long)blockIdx.x % NBH;
   long k0 = ((long)blockIdx.x / NBH) * 64 + w * 16;  // this wave's 16 keys
-  const short* q_p = reinterpret_cast<const short*>(Q) + bh * S * D;
-  const short* k_p = reinterpret_cast<const short*>(K) + bh * S * D;
-  const short* v_p = reinterpret_cast<const short*>(V) + bh * S * D;
+  long bb = bh / H, hh = bh % H;
+  const short* q_p = reinterpret_cast<const short*>(Q) + bb * q_sb
+                     + hh * q_sh;
+  const short* k_p = reinterpret_cast<const short*>(K) + bb * k_sb
+                     + hh * k_sh;
+  const short* v_p = reinterpret_cast<const short*>(V) + bb * v_sb
+                     + hh * v_sh;
   const short* do_p = reinterpret_cast<const short*>(dO) + bh * S * D;
   short* dk_p = reinterpret_cast<short*>(dK) + bh * S * D;
   short* dv_p = reinterpret_cast<short*>(dV) + bh * S * D;
@@ -321,8 +333,8 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
   for (int c = 0; c < NC; ++c) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      kf[c][j] = k_p[krow_a * D + c * 32 + kg * 8 + j];
-      vf[c][j] = v_p[krow_a * D + c * 32 + kg * 8 + j];
+      kf[c][j] = k_p[krow_a * k_ss + c * 32 + kg * 8 + j];
+      vf[c][j] = v_p[krow_a * v_ss + c * 32 + kg * 8 + j];
     }
   }
   // additive mask value of this lane's OWN key rows (kg*4+r)
@@ -346,7 +358,7 @@ attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ Q,
 #pragma unroll
     for (int half = 0; half < NC; ++half) {
       bwd_bf16x8 qv = *reinterpret_cast<const bwd_bf16x8*>(
-          &q_p[qrow * D + scol + half * 8]);
+          &q_p[qrow * q_ss + scol + half * 8]);
       *reinterpret_cast<bwd_bf16x8*>(&Qs[srow][scol + half * 8]) = qv;
 #pragma unroll
       for (int j = 0; j < 8; ++j) QsT[scol + half * 8 + j][srow] = qv[j];

@@ -453,6 +453,15 @@ void psgd_decompress_ef(torch::Tensor flat, torch::Tensor err,
                      Qp.data_ptr<float>(), numel, s, (float)scale);
 }
 
+// Q/K/V may be views of a fused qkv projection: last dim must be
+// contiguous, other dims carried as element strides into the kernels.
+static void attn_check_strided(const torch::Tensor& t, long D,
+                               const char* name) {
+  TORCH_CHECK(t.dim() == 4 && t.size(3) == D && t.stride(3) == 1,
+              name, ": expected [B,H,S,D] with contiguous head dim");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16);
+}
+
 static const float* attn_mask_ptr(const c10::optional<torch::Tensor>& mask,
                                   long B, long S) {
   if (!mask.has_value() || !mask->defined()) return nullptr;
@@ -467,14 +476,16 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        double scale,
                        c10::optional<torch::Tensor> mask = c10::nullopt,
                        double p_drop = 0.0, int64_t seed = 0) {
-  TORCH_CHECK(q.dim() == 4 && (q.size(3) == 64 || q.size(3) == 128),
+  long D = q.size(3);
+  TORCH_CHECK(q.dim() == 4 && (D == 64 || D == 128),
               "attn_fwd expects [B,H,S,64|128]");
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  attn_check_strided(q, D, "q");
+  attn_check_strided(k, D, "k");
+  attn_check_strided(v, D, "v");
   long B = q.size(0), H = q.size(1), S = q.size(2);
   TORCH_CHECK(S % 32 == 0 && S >= 32, "S must be a multiple of 32");
   TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes());
-  auto o = torch::empty_like(q);
+  auto o = torch::empty({B, H, S, D}, q.options());
   // RB=2 (128-row blocks) halves K/V re-read traffic for long sequences;
   // RB=1 keeps more blocks in flight for short ones. 1-D grid with bh as
   // the fast dimension = XCD-locality swizzle (see kernel comment).
@@ -486,8 +497,12 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        reinterpret_cast<__hip_bfloat16*>(k.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(v.data_ptr()),
                        reinterpret_cast<__hip_bfloat16*>(o.data_ptr()),
-                       attn_mask_ptr(mask, B, S), S, H, nbh, (float)scale,
-                       (float)p_drop, (unsigned int)(uint64_t)seed);
+                       attn_mask_ptr(mask, B, S), S, H, nbh,
+                       q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2),
+                       v.stride(0), v.stride(1), v.stride(2),
+                       (float)scale, (float)p_drop,
+                       (unsigned int)(uint64_t)seed);
   };
   if (q.size(3) == 64) {
     if (S >= 256) launch(attn_fwd_kernel<2, 64>, 128);
@@ -515,13 +530,17 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                     c10::optional<torch::Tensor> mask
                                         = c10::nullopt,
                                     double p_drop = 0.0, int64_t seed = 0) {
-  TORCH_CHECK(q.dim() == 4 && (q.size(3) == 64 || q.size(3) == 128));
-  TORCH_CHECK(q.scalar_type() == torch::kBFloat16);
+  long D = q.size(3);
+  TORCH_CHECK(q.dim() == 4 && (D == 64 || D == 128));
+  attn_check_strided(q, D, "q");
+  attn_check_strided(k, D, "k");
+  attn_check_strided(v, D, "v");
+  TORCH_CHECK(o.is_contiguous() && dout.is_contiguous());
   long B = q.size(0), H = q.size(1), S = q.size(2);
   TORCH_CHECK(S % 32 == 0 && S >= 32);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(q);
-  auto dv = torch::empty_like(q);
+  auto dq = torch::empty({B, H, S, D}, q.options());
+  auto dk = torch::empty({B, H, S, D}, q.options());
+  auto dv = torch::empty({B, H, S, D}, q.options());
   auto fopt = q.options().dtype(torch::kFloat32);
   auto Mbuf = torch::empty({B * H * S}, fopt);
   auto Lbuf = torch::empty({B * H * S}, fopt);
@@ -538,23 +557,35 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                        BF16P(q), BF16P(k), BF16P(v), BF16P(o), BF16P(dout),
                        BF16P(dq), Mbuf.data_ptr<float>(),
                        Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
-                       S, H, nbh, (float)scale, (float)p_drop, sd);
+                       S, H, nbh, q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2), v.stride(0),
+                       v.stride(1), v.stride(2), (float)scale,
+                       (float)p_drop, sd);
     hipLaunchKernelGGL(attn_bwd_kv_kernel<64>, grid, dim3(256), 0, st,
                        BF16P(q), BF16P(k), BF16P(v), BF16P(dout), BF16P(dk),
                        BF16P(dv), Mbuf.data_ptr<float>(),
                        Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
-                       S, H, nbh, (float)scale, (float)p_drop, sd);
+                       S, H, nbh, q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2), v.stride(0),
+                       v.stride(1), v.stride(2), (float)scale,
+                       (float)p_drop, sd);
   } else {
     hipLaunchKernelGGL(attn_bwd_q_kernel<128>, grid, dim3(256), 0, st,
                        BF16P(q), BF16P(k), BF16P(v), BF16P(o), BF16P(dout),
                        BF16P(dq), Mbuf.data_ptr<float>(),
                        Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
-                       S, H, nbh, (float)scale, (float)p_drop, sd);
+                       S, H, nbh, q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2), v.stride(0),
+                       v.stride(1), v.stride(2), (float)scale,
+                       (float)p_drop, sd);
     hipLaunchKernelGGL(attn_bwd_kv_kernel<128>, grid, dim3(256), 0, st,
                        BF16P(q), BF16P(k), BF16P(v), BF16P(dout), BF16P(dk),
                        BF16P(dv), Mbuf.data_ptr<float>(),
                        Lbuf.data_ptr<float>(), Dbuf.data_ptr<float>(), mp,
-                       S, H, nbh, (float)scale, (float)p_drop, sd);
+                       S, H, nbh, q.stride(0), q.stride(1), q.stride(2),
+                       k.stride(0), k.stride(1), k.stride(2), v.stride(0),
+                       v.stride(1), v.stride(2), (float)scale,
+                       (float)p_drop, sd);
   }
 #undef BF16P
   return {dq, dk, dv};

@@ -93,8 +93,12 @@ def fused_sdpa(q, k, v, attn_mask: Optional[torch.Tensor] = None,
     if can_use_fused(q, attn_mask, dropout_p):
         mask = _key_padding_mask(attn_mask, q)
         seed = random.getrandbits(31) if dropout_p > 0.0 else 0
-        return FusedAttentionFn.apply(q.contiguous(), k.contiguous(),
-                                      v.contiguous(), scale, mask,
+        # kernels are stride-aware in B/H/S (head dim must be contiguous):
+        # transposed views of the fused qkv projection pass through with
+        # ZERO copies
+        def _ok(t):
+            return t if t.stride(-1) == 1 else t.contiguous()
+        return FusedAttentionFn.apply(_ok(q), _ok(k), _ok(v), scale, mask,
                                       float(dropout_p), seed)
     return torch.nn.functional.scaled_dot_product_attention(
         q, k, v, attn_mask=attn_mask, dropout_p=dropout_p, scale=scale)

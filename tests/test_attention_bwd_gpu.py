@@ -172,3 +172,29 @@ def test_attn_d128_fwd_bwd(B, H, S):
                            ("dv", dv, vf.grad)):
         err = (got.float() - ref).abs().max().item()
         assert err < 1e-1, f"D=128 {name} max err {err}"
+
+
+def test_attn_strided_qkv_views():
+    """Transposed views of a fused qkv projection feed the kernels with
+    ZERO copies — results must match contiguous inputs exactly."""
+    from autodist_amd.ops.fused_attention import fused_sdpa
+    torch.manual_seed(9)
+    B, S, Hh, Dh = 2, 64, 4, 64
+    qkv = torch.randn(B, S, 3, Hh, Dh, device="cuda", dtype=torch.bfloat16)
+    q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
+    assert not q.is_contiguous() and q.stride(-1) == 1
+    qg = q.detach().requires_grad_(True)
+    kg = k.detach().requires_grad_(True)
+    vg = v.detach().requires_grad_(True)
+    out = fused_sdpa(qg, kg, vg)
+    out.float().pow(2).mean().backward()
+    # contiguous reference through the same kernels
+    qc = q.detach().contiguous().requires_grad_(True)
+    kc = k.detach().contiguous().requires_grad_(True)
+    vc = v.detach().contiguous().requires_grad_(True)
+    out_c = fused_sdpa(qc, kc, vc)
+    out_c.float().pow(2).mean().backward()
+    assert torch.equal(out, out_c)
+    assert torch.equal(qg.grad, qc.grad)
+    assert torch.equal(kg.grad, kc.grad)
+    assert torch.equal(vg.grad, vc.grad)
