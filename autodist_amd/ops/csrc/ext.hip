@@ -46,20 +46,16 @@ inline int bn_grid_m(const BNGeom& g, long target_blocks) {
   return (int)std::min(gm, cap);
 }
 
-// max partial rows for the two-stage BN reductions. The 512-block cap of
-// round 1 left only 2 blocks/CU on the big small-C layers and the reduce
-// stage ran at ~1.7 TB/s (measured via tools/bn_microbench.py); 2048
-// blocks (8/CU) saturates HBM3E. Large-C shapes are naturally capped by
-// a partial-matrix byte budget instead.
-#define BN_GM_MAX 2048
+// max partial rows for the two-stage BN reductions (2 blocks per CU).
+// NOTE: raising this to 2048 (8 blocks/CU) was MEASURED SLOWER on every
+// ResNet shape (tools/bn_microbench.py, 2026-09-14): the reduce stage is
+// not occupancy-bound and the larger partial matrix taxes the finalize.
+#define BN_GM_MAX 512
 
 inline int bn_reduce_gm(const BNGeom& g) {
   long rows = (g.M + g.rows_per_blk - 1) / g.rows_per_blk;
-  long gm = (rows + 15) / 16;  // >=16 row-iterations per block
+  long gm = (rows + 31) / 32;
   if (gm < 64) gm = 64;
-  long byte_cap = (2l << 20) / g.C;  // partials <= ~8 MB fp32 per stat
-  if (byte_cap < 256) byte_cap = 256;
-  if (gm > byte_cap) gm = byte_cap;
   if (gm > BN_GM_MAX) gm = BN_GM_MAX;
   if (gm > rows) gm = rows;
   return (int)gm;
