@@ -56,16 +56,7 @@ template <typename T>
 __global__ void bn_fwd_reduce_kernel(const T* __restrict__ x,
                                      float* __restrict__ partial_sum,
                                      float* __restrict__ partial_sq, long M,
-                                     int C, unsigned int* __restrict__ counter,
-                                     const float* __restrict__ weight,
-                                     const float* __restrict__ bias,
-                                     float* __restrict__ running_mean,
-                                     float* __restrict__ running_var,
-                                     float* __restrict__ save_mean,
-                                     float* __restrict__ save_rstd,
-                                     float* __restrict__ scale,
-                                     float* __restrict__ shift, float eps,
-                                     float momentum) {
+                                     int C) {
   int tx_count = min(C / BN_VEC, (int)blockDim.x);
   int tx = threadIdx.x % tx_count;
   int ty = threadIdx.x / tx_count;
@@ -97,42 +88,57 @@ __global__ void bn_fwd_reduce_kernel(const T* __restrict__ x,
     store8(partial_sum + (long)blockIdx.x * C + c0, s);
     store8(partial_sq + (long)blockIdx.x * C + c0, q);
   }
-  // ---- last-block fused finalize: the block that retires the final
-  // partial row sums the partial matrix (hot in L2) and computes the
-  // scale/shift + running stats — saves a ~25 us separate kernel launch
-  // per call (measured; ~2.6 ms/step across ResNet-50's 53 BN layers).
-  __threadfence();
-  __shared__ unsigned int is_last;
-  if (threadIdx.x == 0)
-    is_last = (atomicAdd(counter + blockIdx.y, 1u) + 1 == gridDim.x);
+}
+
+// ------------------------------------------------------------ fwd finalize
+// scale = w*rstd; shift = b - mean*scale; running stats updated in place.
+// block = 64 channels x 4 row-lanes: lanes split the grid_m partial rows so
+// small-C layers still read the partial matrix with thousands of threads.
+#define FIN_CH 64
+#define FIN_LANES 4
+__global__ void bn_fwd_finalize_kernel(const float* __restrict__ partial_sum,
+                                       const float* __restrict__ partial_sq,
+                                       int grid_m,
+                                       const float* __restrict__ weight,
+                                       const float* __restrict__ bias,
+                                       float* __restrict__ running_mean,
+                                       float* __restrict__ running_var,
+                                       float* __restrict__ save_mean,
+                                       float* __restrict__ save_rstd,
+                                       float* __restrict__ scale,
+                                       float* __restrict__ shift, long M,
+                                       int C, float eps, float momentum) {
+  int tx = threadIdx.x % FIN_CH;
+  int ty = threadIdx.x / FIN_CH;
+  int c = blockIdx.x * FIN_CH + tx;
+  if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (int r = ty; r < grid_m; r += FIN_LANES) {
+    s += partial_sum[(long)r * C + c];
+    q += partial_sq[(long)r * C + c];
+  }
+  __shared__ float ls[FIN_LANES * FIN_CH], lq[FIN_LANES * FIN_CH];
+  ls[ty * FIN_CH + tx] = s;
+  lq[ty * FIN_CH + tx] = q;
   __syncthreads();
-  if (!is_last) return;
-  if (threadIdx.x == 0) counter[blockIdx.y] = 0;
-  int cspan = tx_count * BN_VEC;
-  int cbase = blockIdx.y * cspan;
-  int gm = gridDim.x;
-  for (int co = threadIdx.x; co < cspan && cbase + co < C;
-       co += (int)blockDim.x) {
-    int c = cbase + co;
-    float fs = 0.f, fq = 0.f;
-    for (int r = 0; r < gm; ++r) {
-      fs += partial_sum[(long)r * C + c];
-      fq += partial_sq[(long)r * C + c];
-    }
-    float mean = fs / (float)M;
-    float var = fmaxf(fq / (float)M - mean * mean, 0.f);
-    float rstd = rsqrtf(var + eps);
-    float sc = weight[c] * rstd;
-    save_mean[c] = mean;
-    save_rstd[c] = rstd;
-    scale[c] = sc;
-    shift[c] = bias[c] - mean * sc;
-    if (momentum > 0.f) {
-      float unbiased = M > 1 ? var * (float)M / (float)(M - 1) : var;
-      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
-      running_var[c] = (1.f - momentum) * running_var[c]
-                       + momentum * unbiased;
-    }
+  if (ty != 0) return;
+#pragma unroll
+  for (int r = 1; r < FIN_LANES; ++r) {
+    s += ls[r * FIN_CH + tx];
+    q += lq[r * FIN_CH + tx];
+  }
+  float mean = s / (float)M;
+  float var = fmaxf(q / (float)M - mean * mean, 0.f);
+  float rstd = rsqrtf(var + eps);
+  float sc = weight[c] * rstd;
+  save_mean[c] = mean;
+  save_rstd[c] = rstd;
+  scale[c] = sc;
+  shift[c] = bias[c] - mean * sc;
+  if (momentum > 0.f) {
+    float unbiased = M > 1 ? var * (float)M / (float)(M - 1) : var;
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
   }
 }
 
@@ -183,13 +189,7 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const float* __restrict__ save_rstd,
                                      float* __restrict__ partial_dz,
                                      float* __restrict__ partial_dzxh, long M,
-                                     int C, unsigned int* __restrict__ counter,
-                                     const float* __restrict__ weight,
-                                     float* __restrict__ k1,
-                                     float* __restrict__ k2,
-                                     float* __restrict__ k3,
-                                     float* __restrict__ dweight,
-                                     float* __restrict__ dbias) {
+                                     int C) {
   int tx_count = min(C / BN_VEC, (int)blockDim.x);
   int tx = threadIdx.x % tx_count;
   int ty = threadIdx.x / tx_count;
@@ -234,31 +234,46 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     store8(partial_dz + (long)blockIdx.x * C + c0, s);
     store8(partial_dzxh + (long)blockIdx.x * C + c0, t);
   }
-  // ---- last-block fused finalize (see fwd kernel comment)
-  __threadfence();
-  __shared__ unsigned int is_last;
-  if (threadIdx.x == 0)
-    is_last = (atomicAdd(counter + blockIdx.y, 1u) + 1 == gridDim.x);
-  __syncthreads();
-  if (!is_last) return;
-  if (threadIdx.x == 0) counter[blockIdx.y] = 0;
-  int cspan = tx_count * BN_VEC;
-  int cbase = blockIdx.y * cspan;
-  int gm = gridDim.x;
-  for (int co = threadIdx.x; co < cspan && cbase + co < C;
-       co += (int)blockDim.x) {
-    int c = cbase + co;
-    float sdz = 0.f, sdzxh = 0.f;
-    for (int r = 0; r < gm; ++r) {
-      sdz += partial_dz[(long)r * C + c];
-      sdzxh += partial_dzxh[(long)r * C + c];
-    }
-    k1[c] = weight[c] * save_rstd[c];
-    k2[c] = sdz / (float)M;
-    k3[c] = sdzxh / (float)M;
-    dweight[c] = sdzxh;
-    dbias[c] = sdz;
+}
+
+// ------------------------------------------------------------ bwd finalize
+// k1 = w*rstd ; k2 = sum_dz/M ; k3 = sum_dzxh/M ; dweight = sum_dzxh ;
+// dbias = sum_dz
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_dz,
+                                       const float* __restrict__ partial_dzxh,
+                                       int grid_m,
+                                       const float* __restrict__ weight,
+                                       const float* __restrict__ save_rstd,
+                                       float* __restrict__ k1,
+                                       float* __restrict__ k2,
+                                       float* __restrict__ k3,
+                                       float* __restrict__ dweight,
+                                       float* __restrict__ dbias, long M,
+                                       int C) {
+  int tx = threadIdx.x % FIN_CH;
+  int ty = threadIdx.x / FIN_CH;
+  int c = blockIdx.x * FIN_CH + tx;
+  if (c >= C) return;
+  float sdz = 0.f, sdzxh = 0.f;
+  for (int r = ty; r < grid_m; r += FIN_LANES) {
+    sdz += partial_dz[(long)r * C + c];
+    sdzxh += partial_dzxh[(long)r * C + c];
   }
+  __shared__ float ls[FIN_LANES * FIN_CH], lq[FIN_LANES * FIN_CH];
+  ls[ty * FIN_CH + tx] = sdz;
+  lq[ty * FIN_CH + tx] = sdzxh;
+  __syncthreads();
+  if (ty != 0) return;
+#pragma unroll
+  for (int r = 1; r < FIN_LANES; ++r) {
+    sdz += ls[r * FIN_CH + tx];
+    sdzxh += lq[r * FIN_CH + tx];
+  }
+  k1[c] = weight[c] * save_rstd[c];
+  k2[c] = sdz / (float)M;
+  k3[c] = sdzxh / (float)M;
+  dweight[c] = sdzxh;
+  dbias[c] = sdz;
 }
 
 // -------------------------------------------------------------- bwd apply

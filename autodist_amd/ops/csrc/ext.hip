@@ -69,17 +69,15 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
                                         double eps, double momentum,
                                         bool relu,
                                         c10::optional<torch::Tensor> ws) {
-  // ws: persistent per-module workspace [2*BN_GM_MAX + 5, C] fp32:
+  // ws: persistent per-module workspace [2*BN_GM_MAX + 4, C] fp32:
   //   rows [0, GM)               partial sums (per reduce block)
   //   rows [BN_GM_MAX, BN_GM_MAX+GM) partial sumsq
   //   rows 2*BN_GM_MAX + {0,1,2,3}: save_mean, save_rstd, scale, shift
-  //   row  2*BN_GM_MAX + 4: last-block counters (uint storage; MUST start
-  //   zeroed — the fused finalize resets them after each use)
   auto g = bn_geom(x);
   auto fopt = weight.options().dtype(torch::kFloat32);
   torch::Tensor w6 = ws.has_value() ? *ws
-      : torch::zeros({2 * BN_GM_MAX + 5, (long)g.C}, fopt);
-  TORCH_CHECK(w6.size(0) >= 2 * BN_GM_MAX + 5 && w6.size(1) == g.C &&
+      : torch::empty({2 * BN_GM_MAX + 4, (long)g.C}, fopt);
+  TORCH_CHECK(w6.size(0) >= 2 * BN_GM_MAX + 4 && w6.size(1) == g.C &&
               w6.is_contiguous());
   float* wp = w6.data_ptr<float>();
   float* partial_sum = wp;
@@ -88,8 +86,6 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
   auto save_rstd = w6[2 * BN_GM_MAX + 1];
   auto scale = w6[2 * BN_GM_MAX + 2];
   auto shift = w6[2 * BN_GM_MAX + 3];
-  unsigned int* fwd_counter = reinterpret_cast<unsigned int*>(
-      wp + (long)(2 * BN_GM_MAX + 4) * g.C);
   auto y = torch::empty_like(x);
   dim3 block(BLOCK_THREADS);
   int gm = bn_reduce_gm(g);
@@ -100,13 +96,17 @@ std::vector<torch::Tensor> bn_fwd_train(torch::Tensor x, torch::Tensor weight,
 #define BN_FWD_T(T, GET)                                                      \
   {                                                                           \
     hipLaunchKernelGGL((bn_fwd_reduce_kernel<T>), grid_r, block, 0, st,       \
-                       GET(x), partial_sum, partial_sq, g.M, g.C,             \
-                       fwd_counter, weight.data_ptr<float>(),                 \
+                       GET(x), partial_sum, partial_sq, g.M, g.C);            \
+    hipLaunchKernelGGL(bn_fwd_finalize_kernel,                                \
+                       dim3((g.C + FIN_CH - 1) / FIN_CH),                     \
+                       dim3(FIN_CH * FIN_LANES), 0, st, partial_sum,          \
+                       partial_sq, gm,                                        \
+                       weight.data_ptr<float>(),                              \
                        bias.data_ptr<float>(), running_mean.data_ptr<float>(),\
                        running_var.data_ptr<float>(),                         \
                        save_mean.data_ptr<float>(),                           \
                        save_rstd.data_ptr<float>(), scale.data_ptr<float>(),  \
-                       shift.data_ptr<float>(), (float)eps,                   \
+                       shift.data_ptr<float>(), g.M, g.C, (float)eps,         \
                        (float)momentum);                                      \
     if (res.has_value()) {                                                    \
       if (relu)                                                               \
@@ -151,14 +151,12 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                                   bool need_dres,
                                   c10::optional<torch::Tensor> ws) {
   // ws rows: [0,GM) partial_dz, [BN_GM_MAX, BN_GM_MAX+GM) partial_dzxh,
-  //          2*BN_GM_MAX + {0,1,2}: k1, k2, k3; row 2*BN_GM_MAX+3:
-  //          last-block counters (uint storage, self-resetting, must
-  //          start zeroed).
+  //          2*BN_GM_MAX + {0,1,2}: k1, k2, k3.
   auto g = bn_geom(x);
   auto fopt = weight.options().dtype(torch::kFloat32);
   torch::Tensor w5 = ws.has_value() ? *ws
-      : torch::zeros({2 * BN_GM_MAX + 4, (long)g.C}, fopt);
-  TORCH_CHECK(w5.size(0) >= 2 * BN_GM_MAX + 4 && w5.size(1) == g.C &&
+      : torch::empty({2 * BN_GM_MAX + 3, (long)g.C}, fopt);
+  TORCH_CHECK(w5.size(0) >= 2 * BN_GM_MAX + 3 && w5.size(1) == g.C &&
               w5.is_contiguous());
   float* wp = w5.data_ptr<float>();
   float* partial_dz = wp;
@@ -166,8 +164,6 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
   auto k1 = w5[2 * BN_GM_MAX + 0];
   auto k2 = w5[2 * BN_GM_MAX + 1];
   auto k3 = w5[2 * BN_GM_MAX + 2];
-  unsigned int* bwd_counter = reinterpret_cast<unsigned int*>(
-      wp + (long)(2 * BN_GM_MAX + 3) * g.C);
   // fresh allocations: returned to autograd as parameter gradients
   auto dweight = torch::empty({g.C}, fopt);
   auto dbias = torch::empty({g.C}, fopt);
@@ -187,19 +183,22 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor x, torch::Tensor dy,
                          st, GET(x), GET(dy), GET(y),                         \
                          save_mean.data_ptr<float>(),                         \
                          save_rstd.data_ptr<float>(),                         \
-                         partial_dz, partial_dzxh, g.M, g.C, bwd_counter,     \
-                         weight.data_ptr<float>(), k1.data_ptr<float>(),      \
-                         k2.data_ptr<float>(), k3.data_ptr<float>(),          \
-                         dweight.data_ptr<float>(), dbias.data_ptr<float>()); \
+                         partial_dz, partial_dzxh, g.M, g.C);                 \
     else                                                                      \
       hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), grid_r, block, 0,  \
                          st, GET(x), GET(dy), GET(y),                         \
                          save_mean.data_ptr<float>(),                         \
                          save_rstd.data_ptr<float>(),                         \
-                         partial_dz, partial_dzxh, g.M, g.C, bwd_counter,     \
-                         weight.data_ptr<float>(), k1.data_ptr<float>(),      \
-                         k2.data_ptr<float>(), k3.data_ptr<float>(),          \
-                         dweight.data_ptr<float>(), dbias.data_ptr<float>()); \
+                         partial_dz, partial_dzxh, g.M, g.C);                 \
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel,                                \
+                       dim3((g.C + FIN_CH - 1) / FIN_CH),                     \
+                       dim3(FIN_CH * FIN_LANES), 0, st, partial_dz,           \
+                       partial_dzxh, gm,                                      \
+                       weight.data_ptr<float>(),                              \
+                       save_rstd.data_ptr<float>(), k1.data_ptr<float>(),     \
+                       k2.data_ptr<float>(), k3.data_ptr<float>(),            \
+                       dweight.data_ptr<float>(), dbias.data_ptr<float>(),    \
+                       g.M, g.C);                                             \
     T* dres_p = need_dres ? GET(dres) : (T*)nullptr;                          \
     if (relu) {                                                               \
       if (need_dres)                                                          \

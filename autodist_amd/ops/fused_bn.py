@@ -124,13 +124,11 @@ class FusedBatchNorm2d(torch.nn.Module):
             self.num_batches_tracked += 1
             if x.is_cuda and (self._ws is None
                               or self._ws[0].device != x.device):
-                # layouts match BN_GM_MAX=512 in csrc/ext.hip; zeros: the
-                # trailing row holds the fused-finalize last-block
-                # counters which must start at 0
+                # layouts match BN_GM_MAX=512 in csrc/ext.hip
                 self._ws = (
-                    torch.zeros(2 * 512 + 5, self.num_features,
+                    torch.empty(2 * 512 + 4, self.num_features,
                                 device=x.device),
-                    torch.zeros(2 * 512 + 4, self.num_features,
+                    torch.empty(2 * 512 + 3, self.num_features,
                                 device=x.device))
             return fused_bn_train(x, self.weight, self.bias,
                                   self.running_mean, self.running_var,

@@ -46,8 +46,8 @@ def main():
         rm, rv = torch.zeros(C, device=dev), torch.ones(C, device=dev)
         gy = torch.randn_like(x)
 
-        ws = (torch.zeros(2 * 512 + 5, C, device=dev),
-              torch.zeros(2 * 512 + 4, C, device=dev))
+        ws = (torch.empty(2 * 512 + 4, C, device=dev),
+              torch.empty(2 * 512 + 3, C, device=dev))
 
         def fused_fwd():
             with torch.no_grad():
