@@ -115,3 +115,31 @@ def _lm1b_engine_case(rank, world):
 @pytest.mark.integration
 def test_lm1b_sharded_softmax_engine_world2():
     run_distributed(_lm1b_engine_case, world_size=2)
+
+
+def _ckpt_case(rank, world):
+    """state_dict of the sharded LM1B reassembles full tensors; loading a
+    full checkpoint re-shards (SaveSliceInfo semantics)."""
+    import io
+    torch.manual_seed(11)
+    from autodist_amd.models.lm1b import LM1BModel
+    m = LM1BModel(vocab_size=64, emb_dim=16, hidden=32, proj=16,
+                  dropout=0.0, sharded_softmax=True)
+    sd = m.state_dict()
+    assert tuple(sd["emb.weight"].shape) == (64, 16)
+    assert tuple(sd["out.bias"].shape) == (64,)
+    buf = io.BytesIO()
+    torch.save(sd, buf)
+    buf.seek(0)
+    # a re-sharded fresh model loads the full tensors back into its shard
+    torch.manual_seed(99)
+    m2 = LM1BModel(vocab_size=64, emb_dim=16, hidden=32, proj=16,
+                   dropout=0.0, sharded_softmax=True)
+    m2.load_state_dict(torch.load(buf, weights_only=False))
+    assert torch.allclose(m2.emb.shard, m.emb.shard)
+    assert torch.allclose(m2.out.bias, m.out.bias)
+
+
+@pytest.mark.integration
+def test_vocab_parallel_checkpoint_world2():
+    run_distributed(_ckpt_case, world_size=2)
