@@ -505,6 +505,9 @@ torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        (float)scale, (float)p_drop,
                        (unsigned int)(uint64_t)seed);
   };
+  // RB=3 (192-row blocks) for S>=512 was MEASURED SLOWER (0.034 -> 0.048
+  // ms at S=512): occupancy 3 -> 2 waves/SIMD plus partial-tile waste
+  // outweigh the traffic saving. RB=2 is the sweet spot.
   if (q.size(3) == 64) {
     if (S >= 256) launch(attn_fwd_kernel<2, 64>, 128);
     else launch(attn_fwd_kernel<1, 64>, 64);
