@@ -14,6 +14,8 @@ fp32 with bf16 autocast compute.
 import torch
 import torch.nn as nn
 
+from autodist_amd.ops.fused_linear import FusedLinear, fused_linear
+
 
 class BertConfig:
     def __init__(self, vocab_size=30522, hidden=768, layers=12, heads=12,
@@ -46,8 +48,8 @@ class BertSelfAttention(nn.Module):
         super().__init__()
         self.heads = cfg.heads
         self.head_dim = cfg.hidden // cfg.heads
-        self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
-        self.out = nn.Linear(cfg.hidden, cfg.hidden)
+        self.qkv = FusedLinear(cfg.hidden, 3 * cfg.hidden)
+        self.out = FusedLinear(cfg.hidden, cfg.hidden)
         self.dropout = cfg.dropout
 
     def forward(self, x, attn_mask=None):
@@ -69,8 +71,8 @@ class BertLayer(nn.Module):
         from autodist_amd.ops.fused_ln import FusedLayerNorm
         self.attn = BertSelfAttention(cfg)
         self.ln1 = FusedLayerNorm(cfg.hidden, eps=1e-12)
-        self.fc1 = nn.Linear(cfg.hidden, cfg.intermediate)
-        self.fc2 = nn.Linear(cfg.intermediate, cfg.hidden)
+        self.fc1 = FusedLinear(cfg.hidden, cfg.intermediate)
+        self.fc2 = FusedLinear(cfg.intermediate, cfg.hidden)
         self.ln2 = FusedLayerNorm(cfg.hidden, eps=1e-12)
         self.drop = nn.Dropout(cfg.dropout)
 
@@ -124,16 +126,15 @@ class BertForPreTraining(nn.Module):
         super().__init__()
         self.bert = BertModel(cfg, sparse_embeddings)
         from autodist_amd.ops.fused_ln import FusedLayerNorm
-        self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
+        self.mlm_dense = FusedLinear(cfg.hidden, cfg.hidden)
         self.mlm_ln = FusedLayerNorm(cfg.hidden, eps=1e-12)
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
-        self.nsp = nn.Linear(cfg.hidden, 2)
+        self.nsp = FusedLinear(cfg.hidden, 2)
 
     def forward(self, input_ids, token_type_ids=None, attention_mask=None):
         h = self.bert(input_ids, token_type_ids, attention_mask)
         m = self.mlm_ln(torch.nn.functional.gelu(self.mlm_dense(h)))
-        logits = torch.nn.functional.linear(
-            m, self.bert.tok_emb.weight, self.mlm_bias)
+        logits = fused_linear(m, self.bert.tok_emb.weight, self.mlm_bias)
         nsp = self.nsp(h[:, 0])
         return logits, nsp
 

@@ -652,6 +652,24 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor dy, torch::Tensor u,
   return {dx, gb[0], gb[1]};
 }
 
+torch::Tensor col_sum(torch::Tensor x) {
+  TORCH_CHECK(x.dim() == 2 && x.scalar_type() == torch::kBFloat16 &&
+              x.is_contiguous());
+  long N = x.size(0), H = x.size(1);
+  // chunk rows so the grid fills the 8 XCDs even for narrow H
+  long rows_per_chunk = 512;
+  long nchunks = (N + rows_per_chunk - 1) / rows_per_chunk;
+  if (nchunks > 256) { nchunks = 256;
+    rows_per_chunk = (N + nchunks - 1) / nchunks; }
+  auto partials = torch::empty({nchunks, H},
+                               x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(col_sum_kernel, dim3((H + 63) / 64, nchunks),
+                     dim3(256), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(x.data_ptr()),
+                     partials.data_ptr<float>(), N, H, rows_per_chunk);
+  return partials.sum(0);
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B, long cand) {
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32);
   TORCH_CHECK(B.scalar_type() == torch::kBFloat16 && B.numel() == 32 * 16);
@@ -702,6 +720,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "padded M = flat + err (zero tail)");
   m.def("mfma_probe", &mfma_probe,
         "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
+  m.def("col_sum", &col_sum, "bf16 [N,H] column sum -> fp32 [H]");
   m.def("ln_fwd", &ln_fwd,
         "fused bf16 LayerNorm(+residual) forward -> y, u, mean, rstd");
   m.def("ln_bwd", &ln_bwd, "fused bf16 LayerNorm backward -> dx, dgamma, "

@@ -79,3 +79,34 @@ def test_bert_layer_fused_ln_trains():
     out.float().pow(2).mean().backward()
     for p in layer.parameters():
         assert p.grad is None or torch.isfinite(p.grad.float()).all()
+
+
+def test_col_sum_and_fused_linear():
+    from autodist_amd.ops import api
+    torch.manual_seed(5)
+    dy = torch.randn(4096, 768, device="cuda", dtype=torch.bfloat16)
+    ref = dy.float().sum(0)
+    got = api.ext().col_sum(dy)
+    assert (got - ref).abs().max().item() < 0.05 * ref.abs().max().item() + 0.1
+    # FusedLinear fwd+bwd vs nn.Linear under autocast
+    from autodist_amd.ops.fused_linear import FusedLinear
+    lin_ref = torch.nn.Linear(768, 512).cuda()
+    lin = FusedLinear(768, 512).cuda()
+    with torch.no_grad():
+        lin.weight.copy_(lin_ref.weight)
+        lin.bias.copy_(lin_ref.bias)
+    x = torch.randn(8, 32, 768, device="cuda")
+    xr = x.clone().requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    with torch.autocast("cuda", torch.bfloat16):
+        yr = lin_ref(xr)
+        yf = lin(xf)
+    assert torch.allclose(yr, yf, atol=2e-2, rtol=1e-2)
+    g = torch.randn_like(yr)
+    yr.backward(g)
+    yf.backward(g)
+    assert lin_ref.weight.grad.dtype == lin.weight.grad.dtype
+    assert (lin.weight.grad - lin_ref.weight.grad).abs().max().item() < 0.2
+    assert (lin.bias.grad - lin_ref.bias.grad).abs().max().item() < \
+        0.05 * lin_ref.bias.grad.abs().max().item() + 0.2
+    assert (xf.grad - xr.grad).abs().max().item() < 0.1
