@@ -35,8 +35,16 @@ class _LinearFn(torch.autograd.Function):
 
 
 def fused_linear(x, weight, bias):
-    """F.linear with the custom backward when on the bf16 GPU path."""
+    """F.linear with the custom backward when on the bf16 GPU path.
+
+    Disabled under hipGraph stream capture: the composed BERT graph
+    (engine buckets + LN + attention + this backward) replays with a GPU
+    memory fault on the current ROCm stack even though every piece
+    captures fine in isolation (tools/graph_probe.py) — so captured
+    steps record the stock F.linear path and eager/multi-rank steps get
+    the fused bias-gradient kernel."""
     if (x.is_cuda and bias is not None and ops_api.has_gpu_ops()
+            and not torch.cuda.is_current_stream_capturing()
             and (x.dtype == torch.bfloat16
                  or torch.is_autocast_enabled())):
         if x.dtype != torch.bfloat16:
