@@ -11,6 +11,7 @@
 #include "attention.hip"
 #include "attention_bwd.hip"
 #include "ln_ops.hip"
+#include "ce_ops.hip"
 
 namespace {
 
@@ -670,6 +671,38 @@ torch::Tensor col_sum(torch::Tensor x) {
   return partials.sum(0);
 }
 
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits,
+                                  torch::Tensor targets) {
+  TORCH_CHECK(logits.dim() == 2 && logits.scalar_type() == torch::kBFloat16
+              && logits.is_contiguous());
+  TORCH_CHECK(targets.scalar_type() == torch::kLong);
+  long N = logits.size(0), H = logits.size(1);
+  auto fopt = logits.options().dtype(torch::kFloat32);
+  auto loss_rows = torch::empty({N}, fopt);
+  auto m = torch::empty({N}, fopt);
+  auto l2s = torch::empty({N}, fopt);
+  hipLaunchKernelGGL(ce_fwd_kernel, dim3(N), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(logits.data_ptr()),
+                     targets.data_ptr<long>(), loss_rows.data_ptr<float>(),
+                     m.data_ptr<float>(), l2s.data_ptr<float>(), N, H);
+  return {loss_rows, m, l2s};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor targets,
+                     torch::Tensor m, torch::Tensor l2s,
+                     torch::Tensor dloss_rows) {
+  long N = logits.size(0), H = logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  hipLaunchKernelGGL(ce_bwd_kernel, dim3(N), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<__hip_bfloat16*>(logits.data_ptr()),
+                     targets.data_ptr<long>(), m.data_ptr<float>(),
+                     l2s.data_ptr<float>(),
+                     dloss_rows.contiguous().data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(dlogits.data_ptr()),
+                     N, H);
+  return dlogits;
+}
+
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B, long cand) {
   TORCH_CHECK(A.scalar_type() == torch::kBFloat16 && A.numel() == 16 * 32);
   TORCH_CHECK(B.scalar_type() == torch::kBFloat16 && B.numel() == 32 * 16);
@@ -720,6 +753,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "padded M = flat + err (zero tail)");
   m.def("mfma_probe", &mfma_probe,
         "diagnostic: v_mfma_f32_16x16x32_bf16 A/B layout probe");
+  m.def("ce_fwd", &ce_fwd,
+        "fused online-softmax cross-entropy fwd -> loss_rows, m, l2s");
+  m.def("ce_bwd", &ce_bwd, "fused cross-entropy bwd -> dlogits (bf16)");
   m.def("col_sum", &col_sum, "bf16 [N,H] column sum -> fp32 [H]");
   m.def("ln_fwd", &ln_fwd,
         "fused bf16 LayerNorm(+residual) forward -> y, u, mean, rstd");

@@ -29,6 +29,28 @@ import torch.distributed as dist
 from autodist_amd.parallel.partitioner import split_boundaries
 
 
+class _FusedCERows(torch.autograd.Function):
+    """Per-row CE over bf16 logits via the gfx950 online-softmax kernels
+    (ops/csrc/ce_ops.hip): forward streams the logits once keeping only
+    per-row (max, log-sum-exp) — torch's log_softmax materializes and
+    saves the full [N, V] log-probabilities (4 GB at LM1B shape)."""
+
+    @staticmethod
+    def forward(ctx, logits2d, targets):
+        from autodist_amd.ops import api as ops_api
+        loss_rows, m, l2s = ops_api.ext().ce_fwd(logits2d, targets)
+        ctx.save_for_backward(logits2d, targets, m, l2s)
+        return loss_rows
+
+    @staticmethod
+    def backward(ctx, dloss_rows):
+        from autodist_amd.ops import api as ops_api
+        logits2d, targets, m, l2s = ctx.saved_tensors
+        dlogits = ops_api.ext().ce_bwd(logits2d, targets, m, l2s,
+                                       dloss_rows.float())
+        return dlogits, None
+
+
 class _VocabParallelCE(torch.autograd.Function):
     """Mean cross-entropy over vocab-sharded logits."""
 
@@ -140,8 +162,12 @@ class VocabParallelProjection(torch.nn.Module):
             logits = torch.nn.functional.linear(
                 h2, self.weight.to(h2.dtype),
                 self.bias.to(h2.dtype) if self.bias is not None else None)
-            return torch.nn.functional.cross_entropy(logits,
-                                                     targets.reshape(-1))
+            t = targets.reshape(-1)
+            from autodist_amd.ops import api as ops_api
+            if (logits.is_cuda and logits.dtype == torch.bfloat16
+                    and ops_api.has_gpu_ops()):
+                return _FusedCERows.apply(logits.contiguous(), t).mean()
+            return torch.nn.functional.cross_entropy(logits, t)
         return _VocabParallelCE.apply(hidden, self.weight, self.bias,
                                       targets, self.row_start, self.row_end,
                                       self.world_size, self.process_group)

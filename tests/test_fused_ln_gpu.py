@@ -110,3 +110,21 @@ def test_col_sum_and_fused_linear():
     assert (lin.bias.grad - lin_ref.bias.grad).abs().max().item() < \
         0.05 * lin_ref.bias.grad.abs().max().item() + 0.2
     assert (xf.grad - xr.grad).abs().max().item() < 0.1
+
+
+def test_fused_ce_vs_torch():
+    """Online-softmax CE kernels vs F.cross_entropy (loss + input grads)."""
+    from autodist_amd.parallel.vocab_parallel import _FusedCERows
+    torch.manual_seed(6)
+    N, V = 512, 4000
+    logits = torch.randn(N, V, device="cuda", dtype=torch.bfloat16) * 3
+    targets = torch.randint(0, V, (N,), device="cuda")
+    lg = logits.clone().requires_grad_(True)
+    loss = _FusedCERows.apply(lg, targets).mean()
+    loss.backward()
+    lr = logits.float().clone().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lr, targets)
+    ref.backward()
+    assert abs(loss.item() - ref.item()) < 2e-3 * abs(ref.item()) + 1e-3
+    err = (lg.grad.float() - lr.grad).abs().max().item()
+    assert err < 5e-5, f"dlogits err {err}"  # grads are O(1/N)
