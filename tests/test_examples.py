@@ -65,3 +65,9 @@ def test_lm1b_example():
     out = _run(["examples/lm1b/lm1b_train.py", "--small", "--batch-size",
                 "8", "--seq-len", "10", "--steps", "3"])
     assert "words/sec" in out
+
+
+def test_serving_example():
+    out = _run(["examples/serving.py", "--model", "bert_tiny",
+                "--batch", "4", "--seq-len", "32", "--iters", "3"])
+    assert "serving bert_tiny" in out
