@@ -49,7 +49,15 @@ class LM1BModel(nn.Module):
 
     def _hidden(self, tokens, state=None):
         x = self.drop(self.emb(tokens))
-        h, state = self.lstm(x, state)
+        # run the LSTM OUTSIDE autocast in fp32: the decomposed ROCm LSTM
+        # thrashes bf16<->fp32 casts around every pointwise op under
+        # autocast (~170 cast kernels/step measured); a clean fp32 pass is
+        # net faster and more precise. The projection/CE stay bf16.
+        if torch.is_autocast_enabled() and x.is_cuda:
+            with torch.autocast(device_type="cuda", enabled=False):
+                h, state = self.lstm(x.float(), state)
+        else:
+            h, state = self.lstm(x, state)
         return self.drop(h), state
 
     def forward(self, tokens, state=None):
