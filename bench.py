@@ -117,8 +117,9 @@ def build_resnet(args, device, rank, world, use_cuda):
         opt.step()
         return loss
 
+    pretty = "ResNet-50" if args.model == "resnet50" else args.model
     return dict(step=step, engine=engine, items_per_step=B,
-                metric=f"images/sec ResNet-50 {strategy_name}",
+                metric=f"images/sec {pretty} {strategy_name}",
                 unit="images/sec", graph_safe=True,
                 config={"model": args.model, "global_batch": world * B,
                         "seq_len": None, "parallelism": f"dp{world}",
