@@ -10,9 +10,15 @@ GEMMs autograd would run, and db uses the col_sum HIP kernel
 sum). Parameter names match nn.Linear; CPU / non-bf16 paths fall back to
 the stock implementation.
 """
+import os
+
 import torch
 
 from autodist_amd.ops import api as ops_api
+
+# debug override for tools/graph_probe.py: force the fused path INSIDE
+# hipGraph capture to bisect the composed-replay fault
+_FORCE_IN_GRAPH = os.environ.get("AUTODIST_FUSED_LINEAR_IN_GRAPH") == "1"
 
 
 class _LinearFn(torch.autograd.Function):
@@ -44,7 +50,8 @@ def fused_linear(x, weight, bias):
     steps record the stock F.linear path and eager/multi-rank steps get
     the fused bias-gradient kernel."""
     if (x.is_cuda and bias is not None and ops_api.has_gpu_ops()
-            and not torch.cuda.is_current_stream_capturing()
+            and (_FORCE_IN_GRAPH
+                 or not torch.cuda.is_current_stream_capturing())
             and (x.dtype == torch.bfloat16
                  or torch.is_autocast_enabled())):
         if x.dtype != torch.bfloat16:
