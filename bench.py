@@ -50,6 +50,8 @@ def parse_args():
     p.add_argument("--force-collectives", action="store_true",
                    help="execute real RCCL collectives even at world 1 "
                         "(1-GPU hardware validation of the comm path)")
+    p.add_argument("--no-tunableop", action="store_true",
+                   help="disable PYTORCH_TUNABLEOP GEMM algo tuning")
     return p.parse_args()
 
 
@@ -234,6 +236,14 @@ def main():
     args = parse_args()
     if args.force_collectives:
         os.environ["AUTODIST_FORCE_COLLECTIVES"] = "1"
+    if not args.no_tunableop:
+        # hipBLASLt/rocBLAS GEMM algo tuning (per-shape, runs during the
+        # untimed warmup; +1.8% measured on BERT-base). Results go to /tmp
+        # so repo snapshots stay clean.
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME",
+                              "/tmp/tunableop.csv")
     if args.gpus > 1 and "RANK" not in os.environ:
         relaunch_under_torchrun(args)
 
