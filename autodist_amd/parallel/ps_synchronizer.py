@@ -198,16 +198,18 @@ class PSOwnerGroup:
         else:
             # CPU/gloo: async handles so non-owner workers RUN AHEAD within
             # the staleness bound (the c9-verified behavior) — only the
-            # owner blocks, on the reduced gradient it must apply.
-            handles = []
+            # owner blocks, on the reduced gradient it must apply. The
+            # reduce handle rides along in the round so non-owners wait it
+            # before flat_grad is overwritten next step (buffer-reuse
+            # safety); re-waiting a completed handle is a no-op.
             if is_owner:
                 self._reduce_handle.wait()
                 self._grouped_apply(engine)
                 self.flat_stage.copy_(self.flat_master)
             h_bc = dist.broadcast(self.flat_stage, src=self.owner_rank,
                                   group=engine.process_group, async_op=True)
-            handles.append(h_bc)
-            self.rounds.append(PSRound(handle=tuple(handles)))
+            self.rounds.append(
+                PSRound(handle=(self._reduce_handle, h_bc)))
 
     def _wait_round(self, r: PSRound):
         if r.event is not None:

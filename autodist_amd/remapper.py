@@ -54,6 +54,9 @@ class Remapper:
         return value  # scalars / python objects duplicated
 
     def remap_feed_dict(self, feed_dict: Dict[str, Any]) -> Dict[str, Any]:
+        # reset each step: a feed-less step has no uneven split, so its
+        # gradients weight 1/world (not the previous step's fraction)
+        self.batch_fraction = 1.0 / max(self.world_size, 1)
         return {k: self.remap_feed(v) for k, v in (feed_dict or {}).items()}
 
     # -- fetches -----------------------------------------------------------
