@@ -603,7 +603,8 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
   long H = x.size(-1);
   long N = x.numel() / H;
-  TORCH_CHECK(H <= 4096, "ln_fwd supports H <= 4096");
+  TORCH_CHECK(H <= 4096 && H % 2 == 0,
+              "ln_fwd supports even H <= 4096");
   TORCH_CHECK(gamma.scalar_type() == torch::kFloat32 && gamma.is_contiguous());
   auto y = torch::empty_like(x);
   bool has_res = res.has_value() && res->defined();

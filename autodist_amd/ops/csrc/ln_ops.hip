@@ -41,6 +41,7 @@ __device__ __forceinline__ void ln_block_sum2(float& a, float& b,
   __syncthreads();
 }
 
+// paired (bf16x2) loads/stores: H must be even (the module guards)
 __global__ void
 __launch_bounds__(256)
 ln_fwd_kernel(const __hip_bfloat16* __restrict__ x,
@@ -53,21 +54,28 @@ ln_fwd_kernel(const __hip_bfloat16* __restrict__ x,
   __shared__ float lds[8];
   long row = blockIdx.x;
   if (row >= N) return;
-  const __hip_bfloat16* xr = x + row * H;
-  const __hip_bfloat16* rr = res ? res + row * H : nullptr;
-  float v[LN_MAX_PER_THREAD];
-  int nper = (H + 255) >> 8;
+  const __hip_bfloat162* xr =
+      reinterpret_cast<const __hip_bfloat162*>(x + row * H);
+  const __hip_bfloat162* rr = res
+      ? reinterpret_cast<const __hip_bfloat162*>(res + row * H) : nullptr;
+  int H2 = H >> 1;
+  float2 v[LN_MAX_PER_THREAD / 2];
+  int nper = (H2 + 255) >> 8;
   float acc = 0.f, acc2 = 0.f;
   for (int k = 0; k < nper; ++k) {
     int i = threadIdx.x + (k << 8);
-    float u = 0.f;
-    if (i < H) {
-      u = __bfloat162float(xr[i]);
-      if (rr) u += __bfloat162float(rr[i]);
+    float2 u = {0.f, 0.f};
+    if (i < H2) {
+      u = __bfloat1622float2(xr[i]);
+      if (rr) {
+        float2 r2 = __bfloat1622float2(rr[i]);
+        u.x += r2.x;
+        u.y += r2.y;
+      }
     }
     v[k] = u;
-    acc += u;
-    acc2 += u * u;
+    acc += u.x + u.y;
+    acc2 += u.x * u.x + u.y * u.y;
   }
   ln_block_sum2(acc, acc2, lds);
   float mean = acc / H;
@@ -77,14 +85,19 @@ ln_fwd_kernel(const __hip_bfloat16* __restrict__ x,
     mean_out[row] = mean;
     rstd_out[row] = rstd;
   }
-  __hip_bfloat16* yr = y + row * H;
-  __hip_bfloat16* ur = u_out ? u_out + row * H : nullptr;
+  __hip_bfloat162* yr = reinterpret_cast<__hip_bfloat162*>(y + row * H);
+  __hip_bfloat162* ur = u_out
+      ? reinterpret_cast<__hip_bfloat162*>(u_out + row * H) : nullptr;
+  const float2* g2 = reinterpret_cast<const float2*>(gamma);
+  const float2* b2 = reinterpret_cast<const float2*>(beta);
   for (int k = 0; k < nper; ++k) {
     int i = threadIdx.x + (k << 8);
-    if (i < H) {
-      float o = (v[k] - mean) * rstd * gamma[i] + beta[i];
-      yr[i] = __float2bfloat16(o);
-      if (ur) ur[i] = __float2bfloat16(v[k]);
+    if (i < H2) {
+      float2 gg = g2[i], bb = b2[i];
+      float2 o = {(v[k].x - mean) * rstd * gg.x + bb.x,
+                  (v[k].y - mean) * rstd * gg.y + bb.y};
+      yr[i] = __float22bfloat162_rn(o);
+      if (ur) ur[i] = __float22bfloat162_rn(v[k]);
     }
   }
 }
@@ -101,32 +114,42 @@ ln_bwd_dx_kernel(const __hip_bfloat16* __restrict__ dy,
   __shared__ float lds[8];
   long row = blockIdx.x;
   if (row >= N) return;
-  const __hip_bfloat16* dyr = dy + row * H;
-  const __hip_bfloat16* ur = u + row * H;
+  const __hip_bfloat162* dyr =
+      reinterpret_cast<const __hip_bfloat162*>(dy + row * H);
+  const __hip_bfloat162* ur =
+      reinterpret_cast<const __hip_bfloat162*>(u + row * H);
+  const float2* g2 = reinterpret_cast<const float2*>(gamma);
   float mean = mean_in[row], rstd = rstd_in[row];
-  float g[LN_MAX_PER_THREAD], xh[LN_MAX_PER_THREAD];
-  int nper = (H + 255) >> 8;
+  int H2 = H >> 1;
+  float2 g[LN_MAX_PER_THREAD / 2], xh[LN_MAX_PER_THREAD / 2];
+  int nper = (H2 + 255) >> 8;
   float c1 = 0.f, c2 = 0.f;
   for (int k = 0; k < nper; ++k) {
     int i = threadIdx.x + (k << 8);
-    float gv = 0.f, xv = 0.f;
-    if (i < H) {
-      gv = __bfloat162float(dyr[i]) * gamma[i];
-      xv = (__bfloat162float(ur[i]) - mean) * rstd;
+    float2 gv = {0.f, 0.f}, xv = {0.f, 0.f};
+    if (i < H2) {
+      float2 d2 = __bfloat1622float2(dyr[i]);
+      float2 u2 = __bfloat1622float2(ur[i]);
+      float2 gg = g2[i];
+      gv = {d2.x * gg.x, d2.y * gg.y};
+      xv = {(u2.x - mean) * rstd, (u2.y - mean) * rstd};
     }
     g[k] = gv;
     xh[k] = xv;
-    c1 += gv;
-    c2 += gv * xv;
+    c1 += gv.x + gv.y;
+    c2 += gv.x * xv.x + gv.y * xv.y;
   }
   ln_block_sum2(c1, c2, lds);
   c1 /= H;
   c2 /= H;
-  __hip_bfloat16* dxr = dx + row * H;
+  __hip_bfloat162* dxr = reinterpret_cast<__hip_bfloat162*>(dx + row * H);
   for (int k = 0; k < nper; ++k) {
     int i = threadIdx.x + (k << 8);
-    if (i < H)
-      dxr[i] = __float2bfloat16(rstd * (g[k] - c1 - xh[k] * c2));
+    if (i < H2) {
+      float2 o = {rstd * (g[k].x - c1 - xh[k].x * c2),
+                  rstd * (g[k].y - c1 - xh[k].y * c2)};
+      dxr[i] = __float22bfloat162_rn(o);
+    }
   }
 }
 

@@ -55,7 +55,8 @@ class FusedLayerNorm(torch.nn.Module):
             x = x.to(torch.bfloat16)
             if residual is not None:
                 residual = residual.to(torch.bfloat16)
-        if (x.is_cuda and x.dtype == torch.bfloat16 and self.hidden <= 4096
+        if (x.is_cuda and x.dtype == torch.bfloat16
+                and self.hidden <= 4096 and self.hidden % 2 == 0
                 and ops_api.has_gpu_ops()):
             return _FusedLNFn.apply(
                 x.contiguous(),

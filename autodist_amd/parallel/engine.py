@@ -324,7 +324,11 @@ class DistributedEngine:
             if cur is None:
                 continue
             for k, v in group.items():
-                if k != "params" and cur.get(k) != v:
+                if k == "params":
+                    continue
+                if isinstance(v, torch.Tensor):  # capturable-style tensor lr
+                    cur[k] = v
+                elif cur.get(k) != v:
                     cur[k] = v
 
     def set_batch_fraction(self, fraction: Optional[float]):
