@@ -23,6 +23,13 @@ import torch.distributed as dist
 from autodist_amd.parallel.partitioner import split_boundaries
 
 
+def _forced() -> bool:
+    """AUTODIST_FORCE_COLLECTIVES: run the all-to-all exchange even at
+    world 1 (1-GPU RCCL hardware validation; identity routing)."""
+    import os
+    return os.environ.get("AUTODIST_FORCE_COLLECTIVES", "") in ("1", "True")
+
+
 def _alltoallv(parts, world, group, trailing_shape=()):
     """Exchange variable-length dim-0 chunks; parts[r] goes to rank r.
     Returns received chunks (list per source rank)."""
@@ -48,7 +55,7 @@ class _ShardedLookup(torch.autograd.Function):
     @staticmethod
     def forward(ctx, shard_weight, ids, module):
         world = module.world_size
-        if world <= 1:
+        if world <= 1 and not _forced():
             ctx.save_for_backward(ids)
             ctx.module = module
             return _local_gather(shard_weight, ids.reshape(-1)).view(
@@ -79,7 +86,7 @@ class _ShardedLookup(torch.autograd.Function):
     def backward(ctx, grad_out):
         module = ctx.module
         world = module.world_size
-        if world <= 1:
+        if world <= 1 and not _forced():
             (ids,) = ctx.saved_tensors
             g = torch.zeros_like(module.shard)
             _local_scatter_add(g, ids.reshape(-1),
