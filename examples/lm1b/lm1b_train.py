@@ -27,6 +27,9 @@ def main():
     parser.add_argument("--batch-size", type=int, default=128)
     parser.add_argument("--seq-len", type=int, default=20)
     parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--sharded-softmax", action="store_true",
+                        help="vocab-parallel tied projection + sharded CE "
+                             "(parallel/vocab_parallel.py)")
     parser.add_argument("--small", action="store_true",
                         help="10k-vocab model (CPU-sized)")
     args = parser.parse_args()
@@ -35,7 +38,8 @@ def main():
     ad = AutoDist(strategy_builder=getattr(strat, args.autodist_strategy)())
     with ad.scope():
         torch.manual_seed(0)
-        model = lm1b_small() if args.small else lm1b_full()
+        kw = {"sharded_softmax": args.sharded_softmax}
+        model = lm1b_small(**kw) if args.small else lm1b_full(**kw)
         optimizer = torch.optim.Adagrad(model.parameters(), lr=0.01)
     vocab = model.emb.num_embeddings
 
