@@ -80,8 +80,20 @@ class PSOwnerGroup:
         self.numel += n
 
     def allocate(self, engine):
-        self.flat_grad = torch.zeros(self.numel, dtype=self.dtype,
-                                     device=self.device)
+        self._engine = engine
+        # gradient-buffer RING: with staleness k, up to k+1 rounds stay in
+        # flight and a round's reduce may still be READING its grad buffer
+        # while the worker runs ahead — so each step stages into the next
+        # of (depth+2) buffers. A slot only comes back after its round was
+        # consumed (consume_due pops to <= depth and WAITS the round), so
+        # reuse is race-free without blocking the run-ahead (the
+        # c9-verified semantics).
+        depth = self.staleness if self.sync else ASYNC_PS_MAX_DEPTH
+        self._ring = [torch.zeros(self.numel, dtype=self.dtype,
+                                  device=self.device)
+                      for _ in range(depth + 2)]
+        self._slot = 0
+        self.flat_grad = self._ring[0]
         self.flat_stage = torch.empty(self.numel, dtype=self.dtype,
                                       device=self.device)
         is_owner = engine.rank == self.owner_rank
@@ -105,6 +117,8 @@ class PSOwnerGroup:
         self._ready = 0
         self._issued = False
         self._grads_in.clear()
+        self._slot = (self._slot + 1) % len(self._ring)
+        self.flat_grad = self._ring[self._slot]
 
     def copy_grad_in(self, plan, sh, off, n):
         """Stage one shard's gradient into the flat buffer (idempotent per
