@@ -138,6 +138,11 @@ class ShardedEmbedding(torch.nn.Module):
         self.rank = rank if rank is not None else int(
             os.environ.get("RANK", 0))
         self.process_group = process_group
+        if num_embeddings < self.world_size:
+            raise ValueError(
+                f"ShardedEmbedding: {num_embeddings} rows cannot shard "
+                f"across {self.world_size} ranks (use nn.Embedding for "
+                f"tiny tables)")
         bounds = split_boundaries(num_embeddings, self.world_size)
         self.row_start, self.row_end = bounds[self.rank]
         # boundaries for bucketize: end of each shard except the last

@@ -135,6 +135,10 @@ class VocabParallelProjection(torch.nn.Module):
         self.rank = rank if rank is not None else int(
             os.environ.get("RANK", 0))
         self.process_group = process_group
+        if vocab_size < self.world_size:
+            raise ValueError(
+                f"VocabParallelProjection: vocab {vocab_size} cannot shard "
+                f"across {self.world_size} ranks")
         bounds = split_boundaries(vocab_size, self.world_size)
         self.row_start, self.row_end = bounds[self.rank]
         nrows = self.row_end - self.row_start
