@@ -9,6 +9,9 @@ GEMMs autograd would run, and db uses the col_sum HIP kernel
 (ops/csrc/ln_ops.hip — coalesced column-block streaming + small partial
 sum). Parameter names match nn.Linear; CPU / non-bf16 paths fall back to
 the stock implementation.
+
+Reference context: the reference leaves dense layers to TF's runtime
+(SURVEY §2.3); this is hot-path native coverage beyond reference parity.
 """
 import os
 

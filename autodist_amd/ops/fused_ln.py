@@ -9,6 +9,11 @@ residual add in, normalizes with fp32 row statistics, and writes bf16
 (ops/csrc/ln_ops.hip; backward = dx kernel + chunked dgamma/dbeta
 partials). Parameter names match nn.LayerNorm (weight/bias) so state
 dicts are interchangeable.
+
+Reference context: the reference delegates all normalization math to TF
+ops (SURVEY §2.3 — zero native files); hand-written CDNA4 normalization
+on the training hot path is the rebuild's north-star requirement, like
+the fused BN trio (ops/csrc/bn_ops.hip) for the CNN side.
 """
 from typing import Optional
 
